@@ -1,0 +1,123 @@
+"""Data layer: vocab, bucketing sampler properties, padding/masks, synth."""
+
+import os
+import pickle
+
+import numpy as np
+import pytest
+import torch
+
+from wap_amd.data import (
+    Bucket, BucketLoader, Sample, Vocab, load_dict, load_pairs, make_buckets,
+    prepare_batch, shard_buckets, synth_samples, write_synth_dataset,
+)
+from wap_amd.data.vocab import EOS_ID
+
+
+def test_vocab_contract(vocab):
+    assert len(vocab) == 112
+    assert vocab.token_to_id["<eos>"] == 0
+    ids = vocab.encode(r"\frac 1 2")
+    assert len(ids) == 3 and all(i > 0 for i in ids)
+    assert vocab.decode(ids + [EOS_ID, 5]) == r"\frac 1 2"
+
+
+def test_load_dict_rejects_bad_eos(tmp_path):
+    p = tmp_path / "d.txt"
+    p.write_text("a 0\n<eos> 1\n")
+    with pytest.raises(ValueError):
+        load_dict(str(p))
+
+
+def _mk_samples(n=40, seed=0, min_h=16, max_h=80, min_w=16, max_w=160):
+    v = Vocab.default()
+    feats, caps = synth_samples(n, v, seed=seed, min_h=min_h, max_h=max_h,
+                                min_w=min_w, max_w=max_w, min_len=2, max_len=12)
+    return [Sample(u, feats[u], v.encode(caps[u])) for u in feats]
+
+
+def test_bucketing_properties():
+    samples = _mk_samples(60)
+    budget = 40000
+    buckets, dropped = make_buckets(samples, batch_size=8, batch_imagesize=budget,
+                                    maxlen=200, max_imagesize=500000)
+    assert dropped == 0
+    # every sample appears exactly once
+    uids = [s.uid for b in buckets for s in b.samples]
+    assert sorted(uids) == sorted(s.uid for s in samples)
+    for b in buckets:
+        assert len(b) <= 8
+        hmax = max(s.image.shape[0] for s in b.samples)
+        wmax = max(s.image.shape[1] for s in b.samples)
+        if len(b) > 1:  # padded pixel cost respects the budget
+            assert hmax * wmax * len(b) <= budget
+    # deterministic
+    buckets2, _ = make_buckets(samples, batch_size=8, batch_imagesize=budget)
+    assert [[s.uid for s in b.samples] for b in buckets] == \
+           [[s.uid for s in b.samples] for b in buckets2]
+
+
+def test_bucketing_drops_oversize():
+    samples = _mk_samples(10)
+    samples[0] = Sample("big", np.zeros((900, 900), np.uint8), [1, 2])
+    samples[1] = Sample("long", samples[1].image, list(range(1, 110)) * 3)
+    buckets, dropped = make_buckets(samples, maxlen=200, max_imagesize=500000)
+    assert dropped == 2
+    uids = {s.uid for b in buckets for s in b.samples}
+    assert "big" not in uids and "long" not in uids
+
+
+def test_shard_balance_and_equal_steps():
+    samples = _mk_samples(80)
+    buckets, _ = make_buckets(samples, batch_size=4, batch_imagesize=30000)
+    world = 4
+    shards = [shard_buckets(buckets, r, world) for r in range(world)]
+    steps = [len(s) for s in shards]
+    assert len(set(steps)) == 1 and steps[0] > 0  # equal step count (collectives)
+    loads = [sum(b.pixels for b in s) for s in shards]
+    assert max(loads) <= 1.5 * max(min(loads), 1)  # near-balanced pixels
+    # no bucket on two ranks
+    ids = [id(b) for s in shards for b in s]
+    assert len(ids) == len(set(ids))
+
+
+def test_prepare_batch_padding_and_masks():
+    samples = _mk_samples(6)
+    batch = prepare_batch(Bucket(samples))
+    x, xm, y, ym = batch["x"], batch["x_mask"], batch["y"], batch["y_mask"]
+    assert x.shape[0] == 6 and x.shape[1] == 1
+    assert x.min() >= 0 and x.max() <= 1
+    for i, s in enumerate(samples):
+        h, w = s.image.shape
+        assert xm[i, :h, :w].all() and xm[i].sum() == h * w
+        assert (x[i, 0] * (1 - xm[i])).abs().sum() == 0  # zero-padded
+        t = len(s.label)
+        assert y[i, :t].tolist() == s.label
+        assert y[i, t] == EOS_ID
+        assert ym[i].sum() == t + 1
+
+
+def test_loader_epoch_shuffle_deterministic():
+    samples = _mk_samples(30)
+    buckets, _ = make_buckets(samples, batch_size=4, batch_imagesize=30000)
+    l1 = BucketLoader(buckets, seed=3)
+    l2 = BucketLoader(buckets, seed=3)
+    l1.set_epoch(1), l2.set_epoch(1)
+    o1 = [b["uids"] for b in l1]
+    o2 = [b["uids"] for b in l2]
+    assert o1 == o2
+    l2.set_epoch(2)
+    o3 = [b["uids"] for b in l2]
+    assert o1 != o3  # reshuffled across epochs
+
+
+def test_on_disk_roundtrip(tmp_path, vocab):
+    pkl, cap = str(tmp_path / "f.pkl"), str(tmp_path / "c.txt")
+    write_synth_dataset(pkl, cap, 12, vocab, seed=5)
+    with open(pkl, "rb") as f:
+        feats = pickle.load(f)
+    assert all(isinstance(v, np.ndarray) and v.dtype == np.uint8 for v in feats.values())
+    samples = load_pairs(pkl, cap, vocab)
+    assert len(samples) == 12
+    assert all(len(s.label) > 0 for s in samples)
+    assert all(s.image.max() > 0 for s in samples)  # not zero-filled (DVFS rule)

@@ -1,0 +1,213 @@
+"""Data iteration: pkl features + caption file -> pixel-budget buckets.
+
+Capability parity with the reference family's ``data_iterator.py``
+(``dataIterator`` / ``prepare_data``; SURVEY.md §3.1 #3 — file:line
+unverifiable, reference mount empty):
+
+- features pickle: ``{uid(str): np.uint8[H, W]}`` grayscale images,
+- caption file: lines ``"<uid><ws><tok> <tok> ..."``,
+- samples are sorted by image area and packed into buckets under a pixel
+  budget (``batch_imagesize``), dropping samples whose image exceeds
+  ``max_imagesize`` pixels or whose caption exceeds ``maxlen`` tokens,
+- each bucket is zero-padded to a common H×W with an image mask, and the
+  captions padded with an ``y_mask``.
+
+MI355X-first deviations from the TF reference (deliberate, not oversights):
+
+- tensors are batch-major NCHW ``x[B,1,H,W]`` and ``y[B,T]`` (the reference
+  is NHWC / time-major ``y[T,B]``) — NCHW is the fast layout for MIOpen and
+  for our HIP kernels; semantics are carried by the masks.
+- bucketing is rank-aware from day one: ``shard(rank, world)`` deals whole
+  buckets round-robin *after* sorting buckets by total pixels, so every rank
+  gets a near-equal pixel budget per step and RCCL all-reduce never idles on
+  a straggler.
+"""
+
+from __future__ import annotations
+
+import pickle
+import random
+from dataclasses import dataclass, field
+from typing import Dict, Iterator, List, Sequence, Tuple
+
+import numpy as np
+import torch
+
+from .vocab import EOS_ID, Vocab
+
+
+@dataclass
+class Sample:
+    uid: str
+    image: np.ndarray  # uint8 [H, W]
+    label: List[int]   # token ids, no eos appended yet
+
+
+@dataclass
+class Bucket:
+    samples: List[Sample]
+
+    @property
+    def pixels(self) -> int:
+        return sum(int(s.image.shape[0]) * int(s.image.shape[1]) for s in self.samples)
+
+    def __len__(self) -> int:
+        return len(self.samples)
+
+
+def load_pairs(
+    feature_pkl: str,
+    caption_txt: str,
+    vocab: Vocab,
+) -> List[Sample]:
+    """Read the pickle + caption file into aligned samples (uids present in
+    both; caption file order preserved, mirroring the reference)."""
+    with open(feature_pkl, "rb") as f:
+        features: Dict[str, np.ndarray] = pickle.load(f)
+    out: List[Sample] = []
+    with open(caption_txt, "r", encoding="utf-8") as f:
+        for line in f:
+            parts = line.split()
+            if not parts:
+                continue
+            uid, toks = parts[0], parts[1:]
+            if uid not in features:
+                continue
+            out.append(Sample(uid, np.asarray(features[uid], dtype=np.uint8),
+                              [vocab.token_to_id[t] for t in toks]))
+    return out
+
+
+def make_buckets(
+    samples: Sequence[Sample],
+    batch_size: int = 16,
+    batch_imagesize: int = 500000,
+    maxlen: int = 200,
+    max_imagesize: int = 500000,
+) -> Tuple[List[Bucket], int]:
+    """Sort by image area, pack greedily under the pixel budget and the batch
+    size cap; drop over-size/over-length samples.  Returns (buckets,
+    n_dropped).  Deterministic for a given input order."""
+    kept = []
+    dropped = 0
+    for s in samples:
+        h, w = s.image.shape
+        if len(s.label) > maxlen or h * w > max_imagesize:
+            dropped += 1
+            continue
+        kept.append(s)
+    kept.sort(key=lambda s: (s.image.shape[0] * s.image.shape[1], s.uid))
+
+    buckets: List[Bucket] = []
+    cur: List[Sample] = []
+    cur_h = cur_w = 0
+    for s in kept:
+        h, w = s.image.shape
+        # Padded cost of the would-be bucket: every image pays max_H x max_W.
+        new_h, new_w = max(cur_h, h), max(cur_w, w)
+        if cur and (new_h * new_w * (len(cur) + 1) > batch_imagesize
+                    or len(cur) >= batch_size):
+            buckets.append(Bucket(cur))
+            cur, new_h, new_w = [], h, w
+        cur.append(s)
+        cur_h, cur_w = new_h, new_w
+    if cur:
+        buckets.append(Bucket(cur))
+    return buckets, dropped
+
+
+def shard_buckets(buckets: List[Bucket], rank: int, world_size: int) -> List[Bucket]:
+    """Deal whole buckets to ranks so per-rank pixel totals stay balanced:
+    sort by padded pixel cost descending, then greedy-assign each bucket to
+    the currently lightest rank (deterministic)."""
+    if world_size <= 1:
+        return list(buckets)
+    order = sorted(range(len(buckets)),
+                   key=lambda i: (-buckets[i].pixels, i))
+    loads = [0] * world_size
+    mine: List[Bucket] = []
+    counts = [0] * world_size
+    for i in order:
+        r = min(range(world_size), key=lambda j: (loads[j], counts[j], j))
+        loads[r] += buckets[i].pixels
+        counts[r] += 1
+        if r == rank:
+            mine.append(buckets[i])
+    # All ranks must run the same number of steps per epoch (collectives!).
+    n_steps = min(counts) if min(counts) > 0 else 0
+    return mine[:n_steps]
+
+
+def prepare_batch(
+    bucket: Bucket,
+    device: torch.device | str = "cpu",
+    dtype: torch.dtype = torch.float32,
+) -> Dict[str, torch.Tensor]:
+    """Pad one bucket to common H×W / T and build masks.
+
+    Returns ``x[B,1,H,W]`` in [0,1], ``x_mask[B,H,W]`` (1=real pixel),
+    ``y[B,T]`` with <eos> appended, ``y_mask[B,T]`` (1=real token incl. the
+    eos position) — the same information as the reference's
+    ``prepare_data`` but batch-major NCHW.
+    """
+    b = len(bucket.samples)
+    hmax = max(s.image.shape[0] for s in bucket.samples)
+    wmax = max(s.image.shape[1] for s in bucket.samples)
+    tmax = max(len(s.label) for s in bucket.samples) + 1  # +1 for eos
+
+    x = np.zeros((b, 1, hmax, wmax), dtype=np.float32)
+    x_mask = np.zeros((b, hmax, wmax), dtype=np.float32)
+    y = np.full((b, tmax), EOS_ID, dtype=np.int64)
+    y_mask = np.zeros((b, tmax), dtype=np.float32)
+    uids = []
+    for i, s in enumerate(bucket.samples):
+        h, w = s.image.shape
+        x[i, 0, :h, :w] = s.image.astype(np.float32) / 255.0
+        x_mask[i, :h, :w] = 1.0
+        t = len(s.label)
+        y[i, :t] = s.label
+        y_mask[i, : t + 1] = 1.0  # model must predict the eos too
+        uids.append(s.uid)
+
+    dev = torch.device(device)
+    return {
+        "x": torch.from_numpy(x).to(dev, dtype),
+        "x_mask": torch.from_numpy(x_mask).to(dev, dtype),
+        "y": torch.from_numpy(y).to(dev),
+        "y_mask": torch.from_numpy(y_mask).to(dev, dtype),
+        "uids": uids,
+    }
+
+
+class BucketLoader:
+    """Epoch iterator over (rank-local) buckets with seeded reshuffling of
+    bucket order each epoch — the reference reshuffles bucket order per epoch
+    (SURVEY.md §4.1)."""
+
+    def __init__(
+        self,
+        buckets: List[Bucket],
+        rank: int = 0,
+        world_size: int = 1,
+        seed: int = 0,
+        device: torch.device | str = "cpu",
+        dtype: torch.dtype = torch.float32,
+    ):
+        self.all_buckets = buckets
+        self.local = shard_buckets(buckets, rank, world_size)
+        self.seed = seed
+        self.device = device
+        self.dtype = dtype
+        self.epoch = 0
+
+    def set_epoch(self, epoch: int) -> None:
+        self.epoch = epoch
+
+    def __len__(self) -> int:
+        return len(self.local)
+
+    def __iter__(self) -> Iterator[Dict[str, torch.Tensor]]:
+        order = list(range(len(self.local)))
+        random.Random(self.seed * 1_000_003 + self.epoch).shuffle(order)
+        for i in order:
+            yield prepare_batch(self.local[i], self.device, self.dtype)

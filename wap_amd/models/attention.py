@@ -1,0 +1,65 @@
+"""Attender: coverage-based additive attention.
+
+Per decode step t over spatial positions i in 1..L (L = H'*W'):
+
+    beta_t = sum_{l<t} alpha_l                     # coverage map
+    F      = Q * beta_t                            # 11x11 conv, q=512 ch
+    e_ti   = nu . tanh(W_a s_hat + U_a a_i + U_f F_i)
+    alpha  = masked_softmax_i(e)
+    c_t    = sum_i alpha_i a_i
+
+U_a a is constant across steps for one image, so it is precomputed once per
+sequence (``precompute``).  The score + masked softmax + context reduction
+run through ``wap_amd.ops`` (HIP-fused on gfx950).
+"""
+
+from __future__ import annotations
+
+from typing import Tuple
+
+import torch
+from torch import Tensor, nn
+
+from .. import ops
+
+
+class CoverageAttention(nn.Module):
+    def __init__(self, annotation_dim: int = 684, state_dim: int = 256,
+                 attention_dim: int = 512, coverage_ch: int = 512,
+                 coverage_kernel: int = 11):
+        super().__init__()
+        self.W_a = nn.Linear(state_dim, attention_dim, bias=False)
+        self.U_a = nn.Linear(annotation_dim, attention_dim, bias=True)
+        self.conv_Q = nn.Conv2d(1, coverage_ch, coverage_kernel,
+                                padding=coverage_kernel // 2, bias=False)
+        self.U_f = nn.Linear(coverage_ch, attention_dim, bias=True)
+        self.nu = nn.Parameter(torch.zeros(attention_dim))
+        nn.init.normal_(self.nu, std=attention_dim ** -0.5)
+        self.annotation_dim = annotation_dim
+        self.attention_dim = attention_dim
+
+    def precompute(self, a: Tensor) -> Tuple[Tensor, Tensor]:
+        """a[B,D,H,W] -> (a_flat[B,L,D], ua[B,L,A]) done once per sequence."""
+        b, d, h, w = a.shape
+        a_flat = a.flatten(2).transpose(1, 2).contiguous()  # [B,L,D]
+        ua = self.U_a(a_flat)                               # [B,L,A]
+        return a_flat, ua
+
+    def forward(
+        self,
+        s_hat: Tensor,      # [B, n] pre-attention decoder state
+        a_flat: Tensor,     # [B, L, D]
+        ua: Tensor,         # [B, L, A] precomputed U_a a
+        coverage: Tensor,   # [B, H', W'] running sum of alphas
+        mask: Tensor,       # [B, H', W'] feature-grid mask
+    ) -> Tuple[Tensor, Tensor]:
+        """Returns (context[B,D], alpha[B,L])."""
+        b, hp, wp = coverage.shape
+        f = self.conv_Q(coverage.unsqueeze(1))              # [B,q,H',W']
+        f_flat = f.flatten(2).transpose(1, 2)               # [B,L,q]
+        uf = self.U_f(f_flat)                               # [B,L,A]
+        q = self.W_a(s_hat)                                 # [B,A]
+        alpha, ctx = ops.coverage_attention(
+            q, ua, uf, self.nu, a_flat, mask.flatten(1)
+        )
+        return ctx, alpha

@@ -1,0 +1,136 @@
+"""Parser: conditioned two-cell GRU decoder + maxout readout.
+
+Per step t:
+
+    s_hat_t = GRU1(E y_{t-1}, s_{t-1})       # m=256 embedding, n=256 hidden
+    c_t     = Attend(s_hat_t, a, beta_t)     # coverage attention
+    s_t     = GRU2(c_t, s_hat_t)
+    o_t     = W_o maxout_2(W_s s_t + W_c c_t + W_y E y_{t-1})  # 256->128->K
+    s_0     = tanh(W_init . masked_mean_i(a_i))
+    y_0     = zero embedding (start "token")
+
+GRU steps run through wap_amd.ops (fused gate kernel in training, fully
+fused single-launch cell in decode).
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+from torch import Tensor, nn
+
+from .. import ops
+from .attention import CoverageAttention
+
+
+class GRUCellWAP(nn.Module):
+    """GRU cell with explicit packed weights so the HIP fused kernel maps
+    1:1 (gate order r|z|n, PyTorch convention)."""
+
+    def __init__(self, input_dim: int, hidden_dim: int):
+        super().__init__()
+        self.w_x = nn.Parameter(torch.empty(3 * hidden_dim, input_dim))
+        self.w_h = nn.Parameter(torch.empty(3 * hidden_dim, hidden_dim))
+        self.b_x = nn.Parameter(torch.zeros(3 * hidden_dim))
+        self.b_h = nn.Parameter(torch.zeros(3 * hidden_dim))
+        nn.init.xavier_uniform_(self.w_x)
+        nn.init.orthogonal_(self.w_h[:hidden_dim])
+        nn.init.orthogonal_(self.w_h[hidden_dim:2 * hidden_dim])
+        nn.init.orthogonal_(self.w_h[2 * hidden_dim:])
+
+    def forward(self, x: Tensor, h: Tensor) -> Tensor:
+        return ops.gru_cell(x, h, self.w_x, self.w_h, self.b_x, self.b_h)
+
+    @torch.no_grad()
+    def step_fused(self, x: Tensor, h: Tensor) -> Tensor:
+        return ops.fused_gru_cell(x, h, self.w_x, self.w_h, self.b_x, self.b_h)
+
+
+class Parser(nn.Module):
+    def __init__(
+        self,
+        n_tokens: int = 112,
+        annotation_dim: int = 684,
+        embed_dim: int = 256,
+        hidden_dim: int = 256,
+        attention_dim: int = 512,
+        coverage_ch: int = 512,
+        coverage_kernel: int = 11,
+        maxout_pieces: int = 2,
+        readout_dim: int = 128,
+    ):
+        super().__init__()
+        self.embedding = nn.Embedding(n_tokens, embed_dim)
+        self.gru1 = GRUCellWAP(embed_dim, hidden_dim)
+        self.gru2 = GRUCellWAP(annotation_dim, hidden_dim)
+        self.attention = CoverageAttention(
+            annotation_dim, hidden_dim, attention_dim, coverage_ch, coverage_kernel
+        )
+        self.W_init = nn.Linear(annotation_dim, hidden_dim)
+        # readout: maxout over `maxout_pieces` of readout_dim
+        self.W_s = nn.Linear(hidden_dim, maxout_pieces * readout_dim)
+        self.W_c = nn.Linear(annotation_dim, maxout_pieces * readout_dim, bias=False)
+        self.W_y = nn.Linear(embed_dim, maxout_pieces * readout_dim, bias=False)
+        self.W_o = nn.Linear(readout_dim, n_tokens)
+        self.maxout_pieces = maxout_pieces
+        self.readout_dim = readout_dim
+        self.hidden_dim = hidden_dim
+        self.n_tokens = n_tokens
+
+    def init_state(self, a_flat: Tensor, mask_flat: Tensor) -> Tensor:
+        """s_0 = tanh(W_init . masked_mean(a))."""
+        denom = mask_flat.sum(dim=1, keepdim=True).clamp_min(1.0)
+        mean_a = (a_flat * mask_flat.unsqueeze(-1)).sum(dim=1) / denom
+        return torch.tanh(self.W_init(mean_a))
+
+    def readout(self, s: Tensor, c: Tensor, ey: Tensor) -> Tensor:
+        """Maxout readout -> logits[B,K]."""
+        pre = self.W_s(s) + self.W_c(c) + self.W_y(ey)
+        b = pre.shape[0]
+        pre = pre.view(b, self.readout_dim, self.maxout_pieces)
+        return self.W_o(pre.max(dim=2).values)
+
+    def step(
+        self,
+        y_prev_emb: Tensor,   # [B, m]
+        s_prev: Tensor,       # [B, n]
+        a_flat: Tensor,       # [B, L, D]
+        ua: Tensor,           # [B, L, A]
+        coverage: Tensor,     # [B, H', W']
+        mask: Tensor,         # [B, H', W']
+        fused: bool = False,
+    ) -> Tuple[Tensor, Tensor, Tensor, Tensor]:
+        """One decode step -> (logits, s_t, c_t, alpha)."""
+        gru1 = self.gru1.step_fused if fused else self.gru1
+        gru2 = self.gru2.step_fused if fused else self.gru2
+        s_hat = gru1(y_prev_emb, s_prev)
+        c_t, alpha = self.attention(s_hat, a_flat, ua, coverage, mask)
+        s_t = gru2(c_t, s_hat)
+        logits = self.readout(s_t, c_t, y_prev_emb)
+        return logits, s_t, c_t, alpha
+
+    def forward(
+        self,
+        a: Tensor,          # [B, D, H', W'] annotations
+        mask: Tensor,       # [B, H', W']
+        y: Tensor,          # [B, T] target tokens (teacher forcing)
+    ) -> Tensor:
+        """Teacher-forced unroll -> logits[B, T, K]."""
+        b, _, hp, wp = a.shape
+        a_flat, ua = self.attention.precompute(a)
+        mask_flat = mask.flatten(1)
+        s = self.init_state(a_flat, mask_flat)
+        coverage = torch.zeros(b, hp, wp, device=a.device, dtype=a.dtype)
+        emb_prev = torch.zeros(b, self.embedding.embedding_dim,
+                               device=a.device, dtype=a.dtype)
+        logits = []
+        T = y.shape[1]
+        for t in range(T):
+            step_logits, s, _, alpha = self.step(
+                emb_prev, s, a_flat, ua, coverage, mask
+            )
+            logits.append(step_logits)
+            coverage = coverage + alpha.view(b, hp, wp)
+            emb_prev = self.embedding(y[:, t])
+        return torch.stack(logits, dim=1)

@@ -1,0 +1,150 @@
+"""Data-parallel gradient synchronization: bucketed RCCL all-reduce.
+
+The reference has no parallelism at all; this is the
+MI355X-native DP design:
+
+- one process per GPU, ``torch.distributed`` backend "nccl" (= RCCL on
+  ROCm) over xGMI; "gloo" for CPU tests,
+- gradients accumulate DIRECTLY into flat per-bucket buffers (``p.grad`` is
+  a view into the bucket), so there is no pack/unpack copy,
+- buckets are built in reverse parameter order (backward produces grads
+  back-to-front) and each bucket's all-reduce is launched asynchronously
+  the moment its last grad lands — overlapping communication with the rest
+  of backward,
+- default bucket size 8 MB: WAP's grad volume is ~60 MB bf16/fp32 total;
+  xGMI links are ~153 GB/s/link so comm is sub-ms and the win is overlap +
+  few launches, not ring tuning.
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List, Optional
+
+import torch
+import torch.distributed as dist
+from torch import Tensor, nn
+
+
+class Bucket:
+    def __init__(self, params: List[nn.Parameter], buffer: Tensor):
+        self.params = params
+        self.buffer = buffer
+        self.pending = len(params)
+        self.handle = None
+
+
+class BucketedDDP:
+    """Minimal, explicit DDP: call ``zero_grad()`` before backward and
+    ``finish_gradient_sync()`` after backward, before ``optimizer.step()``.
+
+    Works with world_size == 1 (no-op sync) so the same training loop runs
+    everywhere.
+    """
+
+    def __init__(
+        self,
+        model: nn.Module,
+        bucket_bytes: int = 8 << 20,
+        process_group=None,
+    ):
+        self.model = model
+        self.group = process_group
+        self.world_size = dist.get_world_size(process_group) if _dist_ready() else 1
+        self.buckets: List[Bucket] = []
+        self._param_bucket: Dict[int, Bucket] = {}
+        self._hooks = []
+
+        if self.world_size > 1:
+            # Identical start point on every rank.
+            for p in model.parameters():
+                dist.broadcast(p.data, src=0, group=self.group)
+            for b in model.buffers():
+                dist.broadcast(b.data, src=0, group=self.group)
+
+        params = [p for p in model.parameters() if p.requires_grad]
+        # reverse order: backward computes grads for later layers first
+        params = list(reversed(params))
+        cur: List[nn.Parameter] = []
+        cur_bytes = 0
+        groups: List[List[nn.Parameter]] = []
+        for p in params:
+            n = p.numel() * p.element_size()
+            if cur and cur_bytes + n > bucket_bytes:
+                groups.append(cur)
+                cur, cur_bytes = [], 0
+            cur.append(p)
+            cur_bytes += n
+        if cur:
+            groups.append(cur)
+
+        for g in groups:
+            total = sum(p.numel() for p in g)
+            buf = torch.zeros(total, dtype=g[0].dtype, device=g[0].device)
+            off = 0
+            bucket = Bucket(g, buf)
+            for p in g:
+                view = buf[off: off + p.numel()].view_as(p)
+                p.grad = view  # autograd accumulates into this view
+                off += p.numel()
+                self._param_bucket[id(p)] = bucket
+                if self.world_size > 1:
+                    self._hooks.append(
+                        p.register_post_accumulate_grad_hook(self._on_grad_ready)
+                    )
+            self.buckets.append(bucket)
+
+    # -- training-loop API ------------------------------------------------
+
+    def zero_grad(self) -> None:
+        for b in self.buckets:
+            b.buffer.zero_()
+            b.pending = len(b.params)
+            b.handle = None
+
+    def _on_grad_ready(self, p: nn.Parameter) -> None:
+        b = self._param_bucket[id(p)]
+        b.pending -= 1
+        if b.pending == 0:
+            b.handle = dist.all_reduce(
+                b.buffer, op=dist.ReduceOp.SUM, group=self.group, async_op=True
+            )
+
+    def finish_gradient_sync(self) -> None:
+        if self.world_size <= 1:
+            return
+        for b in self.buckets:
+            if b.handle is None:
+                # A param did not receive a grad this step (e.g. frozen
+                # branch): reduce now so ranks stay consistent.
+                b.handle = dist.all_reduce(
+                    b.buffer, op=dist.ReduceOp.SUM, group=self.group, async_op=True
+                )
+        for b in self.buckets:
+            b.handle.wait()
+            b.buffer.div_(self.world_size)
+            b.handle = None
+
+    # convenience passthroughs
+    def __getattr__(self, name):
+        return getattr(self.model, name)
+
+    def __call__(self, *a, **kw):
+        return self.model(*a, **kw)
+
+
+def _dist_ready() -> bool:
+    return dist.is_available() and dist.is_initialized()
+
+
+def init_distributed(device: str = "auto") -> Dict[str, int]:
+    """Initialize torch.distributed from torchrun env vars (no-op for a
+    single process).  Returns {rank, world_size, local_rank}."""
+    import os
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    if world > 1 and not dist.is_initialized():
+        backend = "nccl" if (device != "cpu" and torch.cuda.is_available()) else "gloo"
+        dist.init_process_group(backend=backend)
+    return {"rank": rank, "world_size": world, "local_rank": local_rank}
