@@ -1,0 +1,165 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: WAP-DenseNet training step throughput (+ decode tok/s).
+
+Contract (driver-run):
+    python bench.py --gpus N --steps K --warmup W
+For N>1 the driver launches via torch.distributed.run with one rank per GPU
+over RCCL; this script reads RANK/LOCAL_RANK/WORLD_SIZE from the env.
+
+Measures the BASELINE.json headline: train images/sec on the WAP-DenseNet
+model, batch 16 per GPU of synthetic CROHME-shape (128x512) images, bf16
+autocast, Adadelta + grad clip (the full training step including optimizer),
+weak scaling (per-GPU work fixed).  After the timed region, rank 0 also
+measures beam-10 decode tok/s and reports it under "extra".
+
+Rank 0 prints exactly ONE JSON line.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+from wap_amd import ops
+from wap_amd.data.synth import fixed_shape_batch
+from wap_amd.models import WAPModel
+from wap_amd.parallel import BucketedDDP, init_distributed
+from wap_amd.runtime import beam_decode
+
+
+def main() -> None:
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--batch", type=int, default=16)
+    p.add_argument("--height", type=int, default=128)
+    p.add_argument("--width", type=int, default=512)
+    p.add_argument("--seq-len", type=int, default=32)
+    p.add_argument("--encoder", type=str, default="densenet")
+    p.add_argument("--dtype", type=str, default="bf16")
+    p.add_argument("--kernels", type=str, default="auto")
+    p.add_argument("--beam", type=int, default=10)
+    p.add_argument("--decode-steps", type=int, default=0,
+                   help="decode measurement rounds (default 1 on GPU, 0 on CPU)")
+    p.add_argument("--no-decode", action="store_true")
+    args = p.parse_args()
+
+    ops.set_backend(args.kernels)
+    info = init_distributed("auto")
+    rank, world = info["rank"], info["world_size"]
+    on_gpu = torch.cuda.is_available()
+    if on_gpu:
+        torch.cuda.set_device(info["local_rank"])
+        device = torch.device("cuda", info["local_rank"])
+    else:
+        device = torch.device("cpu")
+    torch.manual_seed(1234 + rank)
+
+    use_bf16 = on_gpu and args.dtype == "bf16"
+
+    def autocast():
+        return torch.autocast("cuda", dtype=torch.bfloat16, enabled=use_bf16)
+
+    model = WAPModel(n_tokens=112, encoder=args.encoder).to(device)
+    ddp = BucketedDDP(model)
+    opt = torch.optim.Adadelta(model.parameters(), lr=1.0, rho=0.95,
+                               eps=1e-8, weight_decay=1e-4)
+
+    x_np, y_np = fixed_shape_batch(args.batch, args.height, args.width,
+                                   args.seq_len, 112, seed=17 + rank)
+    x = torch.from_numpy(x_np).to(device)
+    x_mask = torch.ones(args.batch, args.height, args.width, device=device)
+    y = torch.from_numpy(y_np).to(device)
+    y_mask = torch.ones(args.batch, args.seq_len, device=device)
+
+    def train_step() -> float:
+        ddp.zero_grad()
+        with autocast():
+            out = model(x, x_mask, y, y_mask)
+        out["loss"].backward()
+        ddp.finish_gradient_sync()
+        torch.nn.utils.clip_grad_norm_(model.parameters(), 100.0)
+        opt.step()
+        return out["loss"]
+
+    def barrier_sync():
+        if world > 1:
+            torch.distributed.barrier()
+        if on_gpu:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        loss = train_step()
+    barrier_sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        loss = train_step()
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+    final_loss = float(loss.item())
+
+    # MAX over ranks of the elapsed time (slowest rank defines throughput).
+    if world > 1:
+        t = torch.tensor([elapsed], device=device if on_gpu else "cpu")
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    value = world * args.batch * args.steps / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    extra = {"final_loss": final_loss}
+    decode_rounds = args.decode_steps if args.decode_steps else (1 if on_gpu else 0)
+    if rank == 0 and decode_rounds > 0 and not args.no_decode:
+        model.eval()
+        db = min(args.batch, 8)
+        with autocast():
+            beam_decode(model, x[:db], x_mask[:db], beam=args.beam, maxlen=32)
+        if on_gpu:
+            torch.cuda.synchronize()
+        tok = 0
+        td0 = time.perf_counter()
+        for _ in range(decode_rounds):
+            with autocast():
+                hyps = beam_decode(model, x[:db], x_mask[:db],
+                                   beam=args.beam, maxlen=64)
+            tok += sum(max(len(h), 1) for h in hyps)
+        if on_gpu:
+            torch.cuda.synchronize()
+        td = time.perf_counter() - td0
+        extra["decode_tok_s"] = tok / td
+        extra["decode_beam"] = args.beam
+        extra["decode_batch"] = db
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "train images/sec, WAP-DenseNet CROHME-shape",
+            "value": value,
+            "unit": "images/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if use_bf16 else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": f"wap-{args.encoder}",
+                "global_batch": world * args.batch,
+                "seq_len": args.seq_len,
+                "image": f"{args.height}x{args.width}",
+                "parallelism": f"dp{world}",
+                "kernels": ops.get_backend(),
+            },
+            "extra": extra,
+        }))
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,165 @@
+"""HIP kernels vs the fp32 eager oracle (numerics).
+
+Every test compares the gfx950 kernel output against the plain-PyTorch fp32
+reference of the same op, on random shapes including odd H'xW' and masked
+tails.  Marked gpu: runs on the MI355X box.
+"""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(autouse=True)
+def _hip_backend():
+    from wap_amd import ops
+    from wap_amd.ops import ext
+    assert ext.available(), "HIP extension must be loadable on the GPU box"
+    ops.set_backend("hip")
+    yield
+    ops.set_backend("auto")
+
+
+def _eager(fn, *args):
+    from wap_amd import ops
+    ops.set_backend("eager")
+    try:
+        return fn(*args)
+    finally:
+        ops.set_backend("hip")
+
+
+SHAPES = [(2, 35, 512), (3, 256, 512), (1, 77, 384)]
+
+
+@pytest.mark.parametrize("B,L,A", SHAPES)
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_attention_score_fwd_bwd(B, L, A, dtype):
+    from wap_amd.ops import attention_score
+    torch.manual_seed(0)
+    dev = "cuda"
+    mk = lambda *s: torch.randn(*s, device=dev, dtype=dtype)
+
+    def run(backend_fn):
+        q = mk(B, A).requires_grad_()
+        ua = mk(B, L, A).requires_grad_()
+        uf = mk(B, L, A).requires_grad_()
+        nu = torch.randn(A, device=dev, dtype=torch.float32).requires_grad_()
+        torch.manual_seed(1)
+        e = backend_fn(attention_score, q, ua, uf, nu) \
+            if backend_fn else attention_score(q, ua, uf, nu)
+        gout = torch.randn_like(e)
+        e.backward(gout)
+        return e, q.grad, ua.grad, uf.grad, nu.grad
+
+    torch.manual_seed(2)
+    hip = run(None)
+    torch.manual_seed(2)
+    ref = run(_eager)
+    tol = 2e-5 if dtype == torch.float32 else 3e-2
+    for h, r, name in zip(hip, ref, ["e", "dq", "dua", "duf", "dnu"]):
+        assert torch.allclose(h.float(), r.float(), atol=tol, rtol=tol), \
+            f"{name}: max diff {(h.float()-r.float()).abs().max()}"
+
+
+@pytest.mark.parametrize("B,L,D", [(2, 33, 684), (4, 256, 684), (1, 130, 120)])
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_masked_softmax_context_fwd_bwd(B, L, D, dtype):
+    from wap_amd.ops import masked_softmax_context
+    torch.manual_seed(0)
+    dev = "cuda"
+
+    def run(use_eager):
+        torch.manual_seed(3)
+        e = torch.randn(B, L, device=dev, dtype=torch.float32).requires_grad_()
+        a = torch.randn(B, L, D, device=dev, dtype=dtype).requires_grad_()
+        mask = torch.ones(B, L, device=dev)
+        mask[:, int(0.7 * L):] = 0  # ragged masked tail
+        if use_eager:
+            alpha, ctx = _eager(masked_softmax_context, e, a, mask)
+        else:
+            alpha, ctx = masked_softmax_context(e, a, mask)
+        (alpha.square().sum() + ctx.square().sum()).backward()
+        return alpha, ctx, e.grad, a.grad
+
+    hip = run(False)
+    ref = run(True)
+    tol = 2e-5 if dtype == torch.float32 else 3e-2
+    for h, r, name in zip(hip, ref, ["alpha", "ctx", "de", "da"]):
+        assert torch.allclose(h.float(), r.float(), atol=tol, rtol=tol), \
+            f"{name}: max diff {(h.float()-r.float()).abs().max()}"
+    alpha = hip[0]
+    assert torch.allclose(alpha.sum(1), torch.ones(B, device=dev), atol=1e-5)
+    assert (alpha[:, int(0.7 * L):] == 0).all()
+
+
+@pytest.mark.parametrize("B,n", [(5, 256), (16, 256), (3, 100)])
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_gru_gates_fwd_bwd(B, n, dtype):
+    from wap_amd.ops.functional import _GruGates
+    dev = "cuda"
+
+    def run(use_eager):
+        torch.manual_seed(4)
+        gx = torch.randn(B, 3 * n, device=dev, dtype=dtype).requires_grad_()
+        gh = torch.randn(B, 3 * n, device=dev, dtype=dtype).requires_grad_()
+        h = torch.randn(B, n, device=dev, dtype=dtype).requires_grad_()
+        if use_eager:
+            out = _eager(_GruGates.apply, gx, gh, h)
+        else:
+            out = _GruGates.apply(gx, gh, h)
+        out.square().sum().backward()
+        return out, gx.grad, gh.grad, h.grad
+
+    hip = run(False)
+    ref = run(True)
+    tol = 2e-5 if dtype == torch.float32 else 3e-2
+    for h_, r_, name in zip(hip, ref, ["h_new", "dgx", "dgh", "dh"]):
+        assert torch.allclose(h_.float(), r_.float(), atol=tol, rtol=tol), \
+            f"{name}: max diff {(h_.float()-r_.float()).abs().max()}"
+
+
+@pytest.mark.parametrize("B,n,in_dim", [(10, 256, 256), (16, 256, 684), (7, 100, 50)])
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_fused_gru_cell(B, n, in_dim, dtype):
+    from wap_amd.ops import fused_gru_cell
+    dev = "cuda"
+    torch.manual_seed(5)
+    x = torch.randn(B, in_dim, device=dev, dtype=dtype)
+    h = torch.randn(B, n, device=dev, dtype=dtype)
+    w_x = torch.randn(3 * n, in_dim, device=dev) * 0.05
+    w_h = torch.randn(3 * n, n, device=dev) * 0.05
+    b_x = torch.randn(3 * n, device=dev) * 0.1
+    b_h = torch.randn(3 * n, device=dev) * 0.1
+    out = fused_gru_cell(x, h, w_x, w_h, b_x, b_h)
+    ref = _eager(fused_gru_cell, x.float(), h.float(), w_x, w_h, b_x, b_h)
+    tol = 1e-4 if dtype == torch.float32 else 3e-2
+    assert torch.allclose(out.float(), ref.float(), atol=tol, rtol=tol), \
+        (out.float() - ref.float()).abs().max()
+
+
+def test_fused_gru_cell_no_bias():
+    from wap_amd.ops import fused_gru_cell
+    dev = "cuda"
+    torch.manual_seed(6)
+    x = torch.randn(4, 32, device=dev)
+    h = torch.randn(4, 16, device=dev)
+    w_x = torch.randn(48, 32, device=dev) * 0.1
+    w_h = torch.randn(48, 16, device=dev) * 0.1
+    out = fused_gru_cell(x, h, w_x, w_h, None, None)
+    ref = _eager(fused_gru_cell, x, h, w_x, w_h, None, None)
+    assert torch.allclose(out, ref, atol=1e-4, rtol=1e-4)
+
+
+def test_all_masked_row_is_safe():
+    """A fully-padded image row must produce zero alpha/ctx, not NaN."""
+    from wap_amd.ops import masked_softmax_context
+    dev = "cuda"
+    e = torch.randn(2, 9, device=dev)
+    a = torch.randn(2, 9, 12, device=dev)
+    mask = torch.ones(2, 9, device=dev)
+    mask[1] = 0
+    alpha, ctx = masked_softmax_context(e, a, mask)
+    assert not torch.isnan(alpha).any() and not torch.isnan(ctx).any()
+    assert (alpha[1] == 0).all() and (ctx[1] == 0).all()

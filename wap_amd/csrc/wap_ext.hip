@@ -1,0 +1,560 @@
+// wap_amd gfx950 (CDNA4/MI355X) kernels: fused coverage-attention pieces and
+// GRU steps for Watch-Attend-and-Parse.
+//
+// Written HIP-first for CDNA4 (wave64, LDS-staged, f32 accumulation) — not a
+// port of any CUDA code; the TF reference has no kernels at all (SURVEY.md
+// §3.2), these implement the implicit cuDNN/Eigen workload it launches.
+//
+// Kernels:
+//  - attn_score fwd/bwd: e[b,l] = nu . tanh(q + ua + uf)   
+//  - masked_softmax_context fwd/bwd: masked spatial softmax over L + context
+//    reduction c = sum_l alpha_l a_l, one workgroup per image, alpha staged
+//    in LDS when it fits
+//  - gru_gates fwd/bwd: everything after the two gate GEMMs of a GRU step
+//  - fused_gru_cell: full decode-time GRU step (both gate GEMVs + nonlin)
+//    in ONE launch — the beam-decode loop is latency-bound on launches
+//
+// Numerics: all math accumulates in f32; bf16 tensors are converted at the
+// load/store boundary.  The eager oracle in wap_amd/ops/functional.py uses
+// the same f32 policy so GPU tests compare like for like.
+
+#include <torch/extension.h>
+#include <ATen/Functions.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#include <vector>
+
+#define WAVE 64
+#define DEV __device__ __forceinline__
+
+// ---------------------------------------------------------------- helpers
+
+DEV float ldf(const float* p) { return *p; }
+DEV float ldf(const __hip_bfloat16* p) { return __bfloat162float(*p); }
+DEV void stf(float* p, float v) { *p = v; }
+DEV void stf(__hip_bfloat16* p, float v) { *p = __float2bfloat16(v); }
+
+DEV float wave_sum(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, WAVE);
+  return v;  // valid on lane 0
+}
+
+DEV float wave_max(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v = fmaxf(v, __shfl_down(v, off, WAVE));
+  return v;
+}
+
+// Block-wide reductions, NT threads (NT/64 waves), `red` = 16-float LDS.
+template <int NT>
+DEV float block_sum(float v, float* red) {
+  const int lane = threadIdx.x & (WAVE - 1), w = threadIdx.x >> 6;
+  v = wave_sum(v);
+  if (lane == 0) red[w] = v;
+  __syncthreads();
+  float r = 0.f;
+  if (w == 0) {
+    r = (lane < NT / WAVE) ? red[lane] : 0.f;
+    r = wave_sum(r);
+    if (lane == 0) red[0] = r;
+  }
+  __syncthreads();
+  r = red[0];
+  __syncthreads();
+  return r;
+}
+
+template <int NT>
+DEV float block_max(float v, float* red) {
+  const int lane = threadIdx.x & (WAVE - 1), w = threadIdx.x >> 6;
+  v = wave_max(v);
+  if (lane == 0) red[w] = v;
+  __syncthreads();
+  float r = -3.0e38f;
+  if (w == 0) {
+    r = (lane < NT / WAVE) ? red[lane] : -3.0e38f;
+    r = wave_max(r);
+    if (lane == 0) red[0] = r;
+  }
+  __syncthreads();
+  r = red[0];
+  __syncthreads();
+  return r;
+}
+
+DEV float sigmoidf(float x) { return 1.f / (1.f + __expf(-x)); }
+
+// ------------------------------------------- attention score forward
+// e[b,l] = sum_k nu[k] * tanh(q[b,k] + ua[b,l,k] + uf[b,l,k])
+// One wave per (b,l); q and nu staged in LDS per block.
+
+template <typename T>
+__global__ void attn_score_fwd_kernel(
+    const T* __restrict__ q, const T* __restrict__ ua, const T* __restrict__ uf,
+    const float* __restrict__ nu, float* __restrict__ e, T* __restrict__ tcache,
+    int L, int A) {
+  extern __shared__ float smem[];
+  float* qs = smem;       // [A]
+  float* ns = smem + A;   // [A]
+  const int b = blockIdx.x;
+  for (int k = threadIdx.x; k < A; k += blockDim.x) {
+    qs[k] = ldf(q + (size_t)b * A + k);
+    ns[k] = nu[k];
+  }
+  __syncthreads();
+  const int lane = threadIdx.x & (WAVE - 1), w = threadIdx.x >> 6;
+  const int nw = blockDim.x / WAVE;
+  for (int l = blockIdx.y * nw + w; l < L; l += gridDim.y * nw) {
+    const size_t off = ((size_t)b * L + l) * A;
+    float acc = 0.f;
+    for (int k = lane; k < A; k += WAVE) {
+      const float t = tanhf(qs[k] + ldf(ua + off + k) + ldf(uf + off + k));
+      stf(tcache + off + k, t);
+      acc += ns[k] * t;
+    }
+    acc = wave_sum(acc);
+    if (lane == 0) e[(size_t)b * L + l] = acc;
+  }
+}
+
+// backward: g[b,l,k] = de[b,l] * nu[k] * (1 - t^2)
+//   dua = g (duf identical), dq[b,k] = sum_l g, dnu[k] = sum_{b,l} t * de
+template <typename T>
+__global__ void attn_score_bwd_kernel(
+    const float* __restrict__ de, const T* __restrict__ t,
+    const float* __restrict__ nu, T* __restrict__ dq, T* __restrict__ dua,
+    float* __restrict__ dnu, int L, int A) {
+  const int b = blockIdx.x;
+  for (int k = threadIdx.x; k < A; k += blockDim.x) {
+    const float nk = nu[k];
+    float accq = 0.f, accn = 0.f;
+    for (int l = 0; l < L; ++l) {
+      const size_t off = ((size_t)b * L + l) * A + k;
+      const float tv = ldf(t + off);
+      const float d = de[(size_t)b * L + l];
+      const float g = d * nk * (1.f - tv * tv);
+      stf(dua + off, g);
+      accq += g;
+      accn += tv * d;
+    }
+    stf(dq + (size_t)b * A + k, accq);
+    atomicAdd(dnu + k, accn);
+  }
+}
+
+// --------------------------------- masked softmax over L + context reduce
+// One workgroup (256 threads) per image.  alpha staged in LDS when L fits.
+
+template <typename T, bool LDS_ALPHA>
+__global__ void msc_fwd_kernel(
+    const float* __restrict__ e, const T* __restrict__ a,
+    const float* __restrict__ mask, float* __restrict__ alpha,
+    float* __restrict__ ctx, int L, int D) {
+  extern __shared__ float smem[];
+  float* red = smem;                 // [16]
+  float* alsh = smem + 16;           // [L] when LDS_ALPHA
+  const int b = blockIdx.x;
+  e += (size_t)b * L;
+  mask += (size_t)b * L;
+  alpha += (size_t)b * L;
+  a += (size_t)b * L * D;
+  ctx += (size_t)b * D;
+  float* al = LDS_ALPHA ? alsh : alpha;
+
+  float m = -3.0e38f;
+  for (int l = threadIdx.x; l < L; l += blockDim.x)
+    if (mask[l] > 0.f) m = fmaxf(m, e[l]);
+  m = block_max<256>(m, red);
+
+  float s = 0.f;
+  for (int l = threadIdx.x; l < L; l += blockDim.x) {
+    const float v = mask[l] > 0.f ? __expf(e[l] - m) : 0.f;
+    al[l] = v;
+    s += v;
+  }
+  __syncthreads();
+  s = block_sum<256>(s, red);
+  const float inv = 1.f / fmaxf(s, 1e-12f);
+  for (int l = threadIdx.x; l < L; l += blockDim.x) {
+    const float v = al[l] * inv;
+    al[l] = v;
+    if (LDS_ALPHA) alpha[l] = v;
+  }
+  __syncthreads();
+  // context: lanes own consecutive d (coalesced over a's fastest dim)
+  for (int d = threadIdx.x; d < D; d += blockDim.x) {
+    float acc = 0.f;
+    for (int l = 0; l < L; ++l) acc += al[l] * ldf(a + (size_t)l * D + d);
+    ctx[d] = acc;
+  }
+}
+
+// backward:
+//   t_l  = g_alpha_l + g_ctx . a_l          (staged into de[] as scratch)
+//   de_l = alpha_l * (t_l - sum_m alpha_m t_m)
+//   da[l,d] = alpha_l * g_ctx[d]
+template <typename T>
+__global__ void msc_bwd_kernel(
+    const float* __restrict__ g_alpha, const float* __restrict__ g_ctx,
+    const float* __restrict__ alpha, const T* __restrict__ a,
+    float* __restrict__ de, T* __restrict__ da, int L, int D) {
+  extern __shared__ float smem[];
+  float* red = smem;           // [16]
+  float* gc = smem + 16;       // [D]
+  const int b = blockIdx.x;
+  g_alpha += (size_t)b * L;
+  g_ctx += (size_t)b * D;
+  alpha += (size_t)b * L;
+  a += (size_t)b * L * D;
+  de += (size_t)b * L;
+  da += (size_t)b * L * D;
+
+  for (int d = threadIdx.x; d < D; d += blockDim.x) gc[d] = g_ctx[d];
+  __syncthreads();
+
+  const int lane = threadIdx.x & (WAVE - 1), w = threadIdx.x >> 6;
+  const int nw = blockDim.x / WAVE;
+  for (int l = w; l < L; l += nw) {
+    float acc = 0.f;
+    for (int d = lane; d < D; d += WAVE) acc += gc[d] * ldf(a + (size_t)l * D + d);
+    acc = wave_sum(acc);
+    if (lane == 0) de[l] = g_alpha[l] + acc;  // de used as t-scratch
+  }
+  __syncthreads();
+  float sp = 0.f;
+  for (int l = threadIdx.x; l < L; l += blockDim.x) sp += alpha[l] * de[l];
+  sp = block_sum<256>(sp, red);
+  for (int l = threadIdx.x; l < L; l += blockDim.x)
+    de[l] = alpha[l] * (de[l] - sp);
+  for (size_t i = threadIdx.x; i < (size_t)L * D; i += blockDim.x) {
+    const int l = i / D, d = i % D;
+    stf(da + i, alpha[l] * gc[d]);
+  }
+}
+
+// -------------------------------------------------- GRU gate nonlinearity
+// gates packed [r|z|n]:
+//   r = sig(gxr+ghr); z = sig(gxz+ghz); nn = tanh(gxn + r*ghn)
+//   h' = (1-z)*nn + z*h ; cache = [r|z|nn|ghn]
+
+template <typename T>
+__global__ void gru_gates_fwd_kernel(
+    const T* __restrict__ gx, const T* __restrict__ gh, const T* __restrict__ h,
+    T* __restrict__ h_new, T* __restrict__ cache, long total, int n) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= total) return;
+  const long b = i / n, j = i % n;
+  const size_t o3 = (size_t)b * 3 * n;
+  const float r = sigmoidf(ldf(gx + o3 + j) + ldf(gh + o3 + j));
+  const float z = sigmoidf(ldf(gx + o3 + n + j) + ldf(gh + o3 + n + j));
+  const float ghn = ldf(gh + o3 + 2 * n + j);
+  const float nn = tanhf(ldf(gx + o3 + 2 * n + j) + r * ghn);
+  stf(h_new + i, (1.f - z) * nn + z * ldf(h + i));
+  const size_t o4 = (size_t)b * 4 * n;
+  stf(cache + o4 + j, r);
+  stf(cache + o4 + n + j, z);
+  stf(cache + o4 + 2 * n + j, nn);
+  stf(cache + o4 + 3 * n + j, ghn);
+}
+
+template <typename T>
+__global__ void gru_gates_bwd_kernel(
+    const T* __restrict__ dh_new, const T* __restrict__ h,
+    const T* __restrict__ cache, T* __restrict__ dgx, T* __restrict__ dgh,
+    T* __restrict__ dh, long total, int n) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= total) return;
+  const long b = i / n, j = i % n;
+  const size_t o4 = (size_t)b * 4 * n;
+  const float r = ldf(cache + o4 + j);
+  const float z = ldf(cache + o4 + n + j);
+  const float nn = ldf(cache + o4 + 2 * n + j);
+  const float ghn = ldf(cache + o4 + 3 * n + j);
+  const float g = ldf(dh_new + i);
+  const float hv = ldf(h + i);
+  const float dz = g * (hv - nn) * z * (1.f - z);
+  const float dnn_pre = g * (1.f - z) * (1.f - nn * nn);
+  const float dr = dnn_pre * ghn * r * (1.f - r);
+  const size_t o3 = (size_t)b * 3 * n;
+  stf(dgx + o3 + j, dr);
+  stf(dgx + o3 + n + j, dz);
+  stf(dgx + o3 + 2 * n + j, dnn_pre);
+  stf(dgh + o3 + j, dr);
+  stf(dgh + o3 + n + j, dz);
+  stf(dgh + o3 + 2 * n + j, dnn_pre * r);
+  stf(dh + i, g * z);
+}
+
+// ------------------------------------------ fused decode-time GRU step
+// One thread per (b, j): six dot products + the gate nonlinearity, one
+// launch for the whole cell.  Decode is latency-bound;
+// the weight matrices (<=2 MB) are L2/LLC-resident across beam steps.
+
+template <typename T>
+__global__ void fused_gru_cell_kernel(
+    const T* __restrict__ x, const T* __restrict__ h,
+    const float* __restrict__ w_x, const float* __restrict__ w_h,
+    const float* __restrict__ b_x, const float* __restrict__ b_h,
+    T* __restrict__ h_new, long total, int n, int in_dim) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= total) return;
+  const long b = i / n, j = i % n;
+  const T* xb = x + (size_t)b * in_dim;
+  const T* hb = h + (size_t)b * n;
+  float axr = 0.f, axz = 0.f, axn = 0.f;
+  const float* wr = w_x + (size_t)j * in_dim;
+  const float* wz = w_x + (size_t)(n + j) * in_dim;
+  const float* wn = w_x + (size_t)(2 * n + j) * in_dim;
+  for (int k = 0; k < in_dim; ++k) {
+    const float xv = ldf(xb + k);
+    axr = fmaf(wr[k], xv, axr);
+    axz = fmaf(wz[k], xv, axz);
+    axn = fmaf(wn[k], xv, axn);
+  }
+  float ahr = 0.f, ahz = 0.f, ahn = 0.f;
+  const float* vr = w_h + (size_t)j * n;
+  const float* vz = w_h + (size_t)(n + j) * n;
+  const float* vn = w_h + (size_t)(2 * n + j) * n;
+  for (int k = 0; k < n; ++k) {
+    const float hv = ldf(hb + k);
+    ahr = fmaf(vr[k], hv, ahr);
+    ahz = fmaf(vz[k], hv, ahz);
+    ahn = fmaf(vn[k], hv, ahn);
+  }
+  if (b_x != nullptr) { axr += b_x[j]; axz += b_x[n + j]; axn += b_x[2 * n + j]; }
+  if (b_h != nullptr) { ahr += b_h[j]; ahz += b_h[n + j]; ahn += b_h[2 * n + j]; }
+  const float r = sigmoidf(axr + ahr);
+  const float z = sigmoidf(axz + ahz);
+  const float nn = tanhf(axn + r * ahn);
+  stf(h_new + i, (1.f - z) * nn + z * ldf(hb + j));
+}
+
+// ====================================================================
+// host side
+// ====================================================================
+
+namespace {
+
+inline hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+#define CHECK_GPU(t) \
+  TORCH_CHECK((t).is_cuda(), #t " must be on the GPU"); \
+  TORCH_CHECK((t).is_contiguous(), #t " must be contiguous")
+
+inline bool is_bf16(const torch::Tensor& t) {
+  return t.scalar_type() == at::kBFloat16;
+}
+
+template <typename T>
+const T* cptr(const torch::Tensor& t) {
+  return reinterpret_cast<const T*>(t.data_ptr());
+}
+template <typename T>
+T* ptr(torch::Tensor& t) {
+  return reinterpret_cast<T*>(t.data_ptr());
+}
+
+#define HIP_OK() do { \
+  hipError_t err = hipGetLastError(); \
+  TORCH_CHECK(err == hipSuccess, "HIP kernel launch failed: ", \
+              hipGetErrorString(err)); \
+} while (0)
+
+}  // namespace
+
+std::vector<torch::Tensor> attention_score_fwd(
+    torch::Tensor q, torch::Tensor ua, torch::Tensor uf, torch::Tensor nu) {
+  CHECK_GPU(q); CHECK_GPU(ua); CHECK_GPU(uf); CHECK_GPU(nu);
+  const int B = ua.size(0), L = ua.size(1), A = ua.size(2);
+  TORCH_CHECK(q.sizes() == torch::IntArrayRef({B, A}));
+  TORCH_CHECK(nu.scalar_type() == at::kFloat, "nu must be f32");
+  TORCH_CHECK(q.scalar_type() == ua.scalar_type() &&
+              uf.scalar_type() == ua.scalar_type(), "q/ua/uf dtype mismatch");
+  auto e = torch::empty({B, L}, ua.options().dtype(at::kFloat));
+  auto tcache = torch::empty_like(ua);
+  const dim3 grid(B, std::min((L + 3) / 4, 2048));
+  const int smem = 2 * A * sizeof(float);
+  if (is_bf16(ua)) {
+    hipLaunchKernelGGL(attn_score_fwd_kernel<__hip_bfloat16>, grid, dim3(256),
+                       smem, cur_stream(),
+                       cptr<__hip_bfloat16>(q), cptr<__hip_bfloat16>(ua),
+                       cptr<__hip_bfloat16>(uf), cptr<float>(nu),
+                       ptr<float>(e), ptr<__hip_bfloat16>(tcache), L, A);
+  } else {
+    hipLaunchKernelGGL(attn_score_fwd_kernel<float>, grid, dim3(256),
+                       smem, cur_stream(),
+                       cptr<float>(q), cptr<float>(ua), cptr<float>(uf),
+                       cptr<float>(nu), ptr<float>(e), ptr<float>(tcache), L, A);
+  }
+  HIP_OK();
+  return {e, tcache};
+}
+
+std::vector<torch::Tensor> attention_score_bwd(
+    torch::Tensor de, torch::Tensor t, torch::Tensor nu) {
+  CHECK_GPU(de); CHECK_GPU(t); CHECK_GPU(nu);
+  const int B = t.size(0), L = t.size(1), A = t.size(2);
+  auto dq = torch::empty({B, A}, t.options());
+  auto dua = torch::empty_like(t);
+  auto dnu = torch::zeros({A}, t.options().dtype(at::kFloat));
+  if (is_bf16(t)) {
+    hipLaunchKernelGGL(attn_score_bwd_kernel<__hip_bfloat16>, dim3(B), dim3(256),
+                       0, cur_stream(),
+                       cptr<float>(de), cptr<__hip_bfloat16>(t), cptr<float>(nu),
+                       ptr<__hip_bfloat16>(dq), ptr<__hip_bfloat16>(dua),
+                       ptr<float>(dnu), L, A);
+  } else {
+    hipLaunchKernelGGL(attn_score_bwd_kernel<float>, dim3(B), dim3(256),
+                       0, cur_stream(),
+                       cptr<float>(de), cptr<float>(t), cptr<float>(nu),
+                       ptr<float>(dq), ptr<float>(dua), ptr<float>(dnu), L, A);
+  }
+  HIP_OK();
+  return {dq, dua, dnu};
+}
+
+std::vector<torch::Tensor> masked_softmax_context_fwd(
+    torch::Tensor e, torch::Tensor a, torch::Tensor mask) {
+  CHECK_GPU(e); CHECK_GPU(a); CHECK_GPU(mask);
+  const int B = a.size(0), L = a.size(1), D = a.size(2);
+  TORCH_CHECK(e.scalar_type() == at::kFloat && mask.scalar_type() == at::kFloat,
+              "e/mask must be f32");
+  auto alpha = torch::empty({B, L}, e.options());
+  auto ctx = torch::empty({B, D}, e.options());
+  const bool lds = L <= 16384;
+  const int smem = (16 + (lds ? L : 0)) * sizeof(float);
+#define LAUNCH_MSC(T, LDS) \
+  hipLaunchKernelGGL((msc_fwd_kernel<T, LDS>), dim3(B), dim3(256), smem, \
+                     cur_stream(), cptr<float>(e), cptr<T>(a), \
+                     cptr<float>(mask), ptr<float>(alpha), ptr<float>(ctx), L, D)
+  if (is_bf16(a)) {
+    if (lds) { LAUNCH_MSC(__hip_bfloat16, true); } else { LAUNCH_MSC(__hip_bfloat16, false); }
+  } else {
+    if (lds) { LAUNCH_MSC(float, true); } else { LAUNCH_MSC(float, false); }
+  }
+#undef LAUNCH_MSC
+  HIP_OK();
+  return {alpha, ctx};
+}
+
+std::vector<torch::Tensor> masked_softmax_context_bwd(
+    torch::Tensor g_alpha, torch::Tensor g_ctx, torch::Tensor alpha,
+    torch::Tensor a) {
+  CHECK_GPU(g_alpha); CHECK_GPU(g_ctx); CHECK_GPU(alpha); CHECK_GPU(a);
+  const int B = a.size(0), L = a.size(1), D = a.size(2);
+  auto de = torch::empty({B, L}, g_alpha.options().dtype(at::kFloat));
+  auto da = torch::empty_like(a);
+  const int smem = (16 + D) * sizeof(float);
+  if (is_bf16(a)) {
+    hipLaunchKernelGGL(msc_bwd_kernel<__hip_bfloat16>, dim3(B), dim3(256), smem,
+                       cur_stream(), cptr<float>(g_alpha), cptr<float>(g_ctx),
+                       cptr<float>(alpha), cptr<__hip_bfloat16>(a),
+                       ptr<float>(de), ptr<__hip_bfloat16>(da), L, D);
+  } else {
+    hipLaunchKernelGGL(msc_bwd_kernel<float>, dim3(B), dim3(256), smem,
+                       cur_stream(), cptr<float>(g_alpha), cptr<float>(g_ctx),
+                       cptr<float>(alpha), cptr<float>(a),
+                       ptr<float>(de), ptr<float>(da), L, D);
+  }
+  HIP_OK();
+  return {de, da};
+}
+
+std::vector<torch::Tensor> gru_gates_fwd(
+    torch::Tensor gx, torch::Tensor gh, torch::Tensor h) {
+  CHECK_GPU(gx); CHECK_GPU(gh); CHECK_GPU(h);
+  const long B = h.size(0);
+  const int n = h.size(1);
+  const long total = B * n;
+  auto h_new = torch::empty_like(h);
+  auto cache = torch::empty({B, 4L * n}, h.options());
+  const int blocks = (total + 255) / 256;
+  if (is_bf16(h)) {
+    hipLaunchKernelGGL(gru_gates_fwd_kernel<__hip_bfloat16>, dim3(blocks),
+                       dim3(256), 0, cur_stream(),
+                       cptr<__hip_bfloat16>(gx), cptr<__hip_bfloat16>(gh),
+                       cptr<__hip_bfloat16>(h), ptr<__hip_bfloat16>(h_new),
+                       ptr<__hip_bfloat16>(cache), total, n);
+  } else {
+    hipLaunchKernelGGL(gru_gates_fwd_kernel<float>, dim3(blocks), dim3(256),
+                       0, cur_stream(),
+                       cptr<float>(gx), cptr<float>(gh), cptr<float>(h),
+                       ptr<float>(h_new), ptr<float>(cache), total, n);
+  }
+  HIP_OK();
+  return {h_new, cache};
+}
+
+std::vector<torch::Tensor> gru_gates_bwd(
+    torch::Tensor dh_new, torch::Tensor h, torch::Tensor cache) {
+  CHECK_GPU(dh_new); CHECK_GPU(h); CHECK_GPU(cache);
+  const long B = h.size(0);
+  const int n = h.size(1);
+  const long total = B * n;
+  auto dgx = torch::empty({B, 3L * n}, h.options());
+  auto dgh = torch::empty({B, 3L * n}, h.options());
+  auto dh = torch::empty_like(h);
+  const int blocks = (total + 255) / 256;
+  if (is_bf16(h)) {
+    hipLaunchKernelGGL(gru_gates_bwd_kernel<__hip_bfloat16>, dim3(blocks),
+                       dim3(256), 0, cur_stream(),
+                       cptr<__hip_bfloat16>(dh_new), cptr<__hip_bfloat16>(h),
+                       cptr<__hip_bfloat16>(cache), ptr<__hip_bfloat16>(dgx),
+                       ptr<__hip_bfloat16>(dgh), ptr<__hip_bfloat16>(dh),
+                       total, n);
+  } else {
+    hipLaunchKernelGGL(gru_gates_bwd_kernel<float>, dim3(blocks), dim3(256),
+                       0, cur_stream(),
+                       cptr<float>(dh_new), cptr<float>(h), cptr<float>(cache),
+                       ptr<float>(dgx), ptr<float>(dgh), ptr<float>(dh),
+                       total, n);
+  }
+  HIP_OK();
+  return {dgx, dgh, dh};
+}
+
+torch::Tensor fused_gru_cell(
+    torch::Tensor x, torch::Tensor h, torch::Tensor w_x, torch::Tensor w_h,
+    c10::optional<torch::Tensor> b_x, c10::optional<torch::Tensor> b_h) {
+  CHECK_GPU(x); CHECK_GPU(h); CHECK_GPU(w_x); CHECK_GPU(w_h);
+  TORCH_CHECK(w_x.scalar_type() == at::kFloat && w_h.scalar_type() == at::kFloat,
+              "weights must be f32");
+  const long B = h.size(0);
+  const int n = h.size(1), in_dim = x.size(1);
+  const long total = B * n;
+  auto h_new = torch::empty_like(h);
+  const float* bx = b_x.has_value() ? cptr<float>(*b_x) : nullptr;
+  const float* bh = b_h.has_value() ? cptr<float>(*b_h) : nullptr;
+  const int blocks = (total + 255) / 256;
+  if (is_bf16(h)) {
+    hipLaunchKernelGGL(fused_gru_cell_kernel<__hip_bfloat16>, dim3(blocks),
+                       dim3(256), 0, cur_stream(),
+                       cptr<__hip_bfloat16>(x), cptr<__hip_bfloat16>(h),
+                       cptr<float>(w_x), cptr<float>(w_h), bx, bh,
+                       ptr<__hip_bfloat16>(h_new), total, n, in_dim);
+  } else {
+    hipLaunchKernelGGL(fused_gru_cell_kernel<float>, dim3(blocks), dim3(256),
+                       0, cur_stream(),
+                       cptr<float>(x), cptr<float>(h),
+                       cptr<float>(w_x), cptr<float>(w_h), bx, bh,
+                       ptr<float>(h_new), total, n, in_dim);
+  }
+  HIP_OK();
+  return h_new;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("attention_score_fwd", &attention_score_fwd);
+  m.def("attention_score_bwd", &attention_score_bwd);
+  m.def("masked_softmax_context_fwd", &masked_softmax_context_fwd);
+  m.def("masked_softmax_context_bwd", &masked_softmax_context_bwd);
+  m.def("gru_gates_fwd", &gru_gates_fwd);
+  m.def("gru_gates_bwd", &gru_gates_bwd);
+  m.def("fused_gru_cell", &fused_gru_cell);
+  m.attr("arch") = "gfx950";
+}

@@ -9,8 +9,12 @@ The three hot paths identified in SURVEY.md §3.2:
 - ``gru_cell`` / ``fused_gru_cell`` — the conditioned-GRU step (gate GEMMs +
   sigmoid/tanh blend; fused single-launch form for latency-bound decode).
 
-Eager forms are the fp32 numerical oracle the HIP kernels are tested
-against (tests/test_kernels_gpu.py).
+Dtype policy (identical for eager and HIP so A/B and numerics tests are
+like-for-like): big per-position tensors (ua/uf/a, GRU gates) stay in their
+incoming dtype (bf16 under autocast); every reduction/softmax/nonlinearity
+accumulates in f32; attention scores ``e``, ``alpha`` and the context are
+f32.  Backward returns grads in each input's original dtype (autograd
+contract).
 """
 
 from __future__ import annotations
@@ -22,48 +26,51 @@ from torch import Tensor
 
 from . import use_hip
 
+def _up(t: Tensor) -> Tensor:
+    """Upcast reduced-precision tensors to f32; leave f32/f64 untouched
+    (keeps gradcheck in double while bf16 runs accumulate in f32)."""
+    return t.float() if t.dtype in (torch.bfloat16, torch.float16) else t
+
+
 
 # --------------------------------------------------------------------------
 # attention score: e[b,l] = sum_k nu[k] * tanh(q[b,k] + ua[b,l,k] + uf[b,l,k])
 # --------------------------------------------------------------------------
 
-def _attention_score_eager(q: Tensor, ua: Tensor, uf: Tensor, nu: Tensor) -> Tensor:
-    t = torch.tanh(q.unsqueeze(1) + ua + uf)
-    return t @ nu
-
-
 class _AttentionScore(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q: Tensor, ua: Tensor, uf: Tensor, nu: Tensor) -> Tensor:
-        if use_hip(q):
+        q_ = q.to(ua.dtype).contiguous()
+        uf_ = uf.to(ua.dtype).contiguous()
+        nu_ = _up(nu).contiguous()
+        if use_hip(ua):
             from . import ext
-            e, tanh_cache = ext.require().attention_score_fwd(
-                q.contiguous(), ua.contiguous(), uf.contiguous(), nu.contiguous()
-            )
+            e, t = ext.require().attention_score_fwd(q_, ua.contiguous(), uf_, nu_)
         else:
-            tanh_cache = torch.tanh(q.unsqueeze(1) + ua + uf)
-            e = tanh_cache @ nu
-        ctx.save_for_backward(tanh_cache, nu)
+            t = torch.tanh(q_.unsqueeze(1) + ua + uf_)
+            e = _up(t) @ nu_
+        ctx.save_for_backward(t, nu_)
+        ctx.in_dtypes = (q.dtype, ua.dtype, uf.dtype, nu.dtype)
         return e
 
     @staticmethod
     def backward(ctx, de: Tensor):
         t, nu = ctx.saved_tensors
-        de = de.contiguous()
-        if use_hip(de):
+        qd, uad, ufd, nud = ctx.in_dtypes
+        de = _up(de).contiguous()
+        if use_hip(t):
             from . import ext
             dq, dua, dnu = ext.require().attention_score_bwd(de, t, nu)
-            # duf == dua element-wise (same broadcast shape)
-            return dq, dua, dua.clone(), dnu
-        # d/dx tanh = 1 - t^2; chain through the nu-dot
-        g = de.unsqueeze(-1) * nu * (1.0 - t * t)   # [B,L,A]
+            return dq.to(qd), dua.to(uad), dua.clone().to(ufd), dnu.to(nud)
+        tf = _up(t)
+        g = de.unsqueeze(-1) * nu * (1.0 - tf * tf)   # [B,L,A] f32
         dq = g.sum(dim=1)
-        dnu = (t * de.unsqueeze(-1)).sum(dim=(0, 1))
-        return dq, g, g.clone(), dnu
+        dnu = (tf * de.unsqueeze(-1)).sum(dim=(0, 1))
+        return dq.to(qd), g.to(uad), g.to(ufd), dnu.to(nud)
 
 
 def attention_score(q: Tensor, ua: Tensor, uf: Tensor, nu: Tensor) -> Tensor:
-    """e[B,L] = nu . tanh(q[B,A] + ua[B,L,A] + uf[B,L,A])."""
+    """e[B,L] (f32) = nu . tanh(q[B,A] + ua[B,L,A] + uf[B,L,A])."""
     return _AttentionScore.apply(q, ua, uf, nu)
 
 
@@ -74,46 +81,50 @@ def attention_score(q: Tensor, ua: Tensor, uf: Tensor, nu: Tensor) -> Tensor:
 def _masked_softmax_context_eager(
     e: Tensor, a: Tensor, mask: Tensor
 ) -> Tuple[Tensor, Tensor]:
-    neg = torch.finfo(e.dtype).min
+    neg = torch.finfo(torch.float32).min
     e = e.masked_fill(mask <= 0, neg)
     alpha = torch.softmax(e, dim=1) * (mask > 0)
     alpha = alpha / alpha.sum(dim=1, keepdim=True).clamp_min(1e-12)
-    ctxv = torch.einsum("bl,bld->bd", alpha, a)
+    ctxv = torch.einsum("bl,bld->bd", alpha, _up(a))
     return alpha, ctxv
 
 
 class _MaskedSoftmaxContext(torch.autograd.Function):
     @staticmethod
     def forward(ctx, e: Tensor, a: Tensor, mask: Tensor):
-        if use_hip(e):
+        e_ = _up(e).contiguous()
+        mask_ = _up(mask).contiguous()
+        if use_hip(a):
             from . import ext
             alpha, ctxv = ext.require().masked_softmax_context_fwd(
-                e.contiguous(), a.contiguous(), mask.contiguous()
+                e_, a.contiguous(), mask_
             )
         else:
-            alpha, ctxv = _masked_softmax_context_eager(e, a, mask)
+            alpha, ctxv = _masked_softmax_context_eager(e_, a, mask_)
         ctx.save_for_backward(alpha, a)
+        ctx.in_dtypes = (e.dtype, a.dtype)
         return alpha, ctxv
 
     @staticmethod
     def backward(ctx, g_alpha: Tensor, g_ctx: Tensor):
         alpha, a = ctx.saved_tensors
-        g_alpha = g_alpha.contiguous()
-        g_ctx = g_ctx.contiguous()
-        if use_hip(alpha):
+        ed, ad = ctx.in_dtypes
+        g_alpha = _up(g_alpha).contiguous()
+        g_ctx = _up(g_ctx).contiguous()
+        if use_hip(a):
             from . import ext
             de, da = ext.require().masked_softmax_context_bwd(g_alpha, g_ctx, alpha, a)
-            return de, da, None
+            return de.to(ed), da, None
         # t_l = g_alpha_l + g_ctx . a_l ; de_l = alpha_l (t_l - sum_m alpha_m t_m)
-        t = g_alpha + torch.einsum("bd,bld->bl", g_ctx, a)
+        t = g_alpha + torch.einsum("bd,bld->bl", g_ctx, _up(a))
         de = alpha * (t - (alpha * t).sum(dim=1, keepdim=True))
         da = alpha.unsqueeze(-1) * g_ctx.unsqueeze(1)
-        return de, da, None
+        return de.to(ed), da.to(ad), None
 
 
 def masked_softmax_context(e: Tensor, a: Tensor, mask: Tensor) -> Tuple[Tensor, Tensor]:
-    """(alpha[B,L], ctx[B,D]) from scores e[B,L], annotations a[B,L,D],
-    mask[B,L] (1 = valid position)."""
+    """(alpha[B,L] f32, ctx[B,D] f32) from scores e[B,L], annotations
+    a[B,L,D], mask[B,L] (1 = valid position)."""
     return _MaskedSoftmaxContext.apply(e, a, mask)
 
 
@@ -134,7 +145,7 @@ def gru_cell(
     w_x: Tensor, w_h: Tensor,
     b_x: Optional[Tensor] = None, b_h: Optional[Tensor] = None,
 ) -> Tensor:
-    """One GRU step, differentiable (training path; GEMMs go to rocBLAS).
+    """One GRU step, differentiable (training path; gate GEMMs -> rocBLAS).
 
     ``w_x``[3n, in], ``w_h``[3n, n]; gates packed [r|z|n]::
 
@@ -153,41 +164,48 @@ class _GruGates(torch.autograd.Function):
 
     @staticmethod
     def forward(ctx, gx: Tensor, gh: Tensor, h: Tensor) -> Tensor:
-        if use_hip(gx):
+        common = torch.promote_types(torch.promote_types(gx.dtype, gh.dtype), h.dtype)
+        gx_ = gx.to(common).contiguous()
+        gh_ = gh.to(common).contiguous()
+        h_ = h.to(common).contiguous()
+        if use_hip(gx_):
             from . import ext
-            h_new, cache = ext.require().gru_gates_fwd(
-                gx.contiguous(), gh.contiguous(), h.contiguous()
-            )
-            ctx.save_for_backward(h, cache)
-            ctx.hip = True
-            return h_new
-        n = h.shape[-1]
-        r = torch.sigmoid(gx[..., :n] + gh[..., :n])
-        z = torch.sigmoid(gx[..., n:2*n] + gh[..., n:2*n])
-        nn = torch.tanh(gx[..., 2*n:] + r * gh[..., 2*n:])
-        h_new = (1.0 - z) * nn + z * h
-        cache = torch.cat([r, z, nn, gh[..., 2*n:]], dim=-1)
-        ctx.save_for_backward(h, cache)
-        ctx.hip = False
+            h_new, cache = ext.require().gru_gates_fwd(gx_, gh_, h_)
+            hip = True
+        else:
+            n = h_.shape[-1]
+            gxf, ghf, hf = _up(gx_), _up(gh_), _up(h_)
+            r = torch.sigmoid(gxf[..., :n] + ghf[..., :n])
+            z = torch.sigmoid(gxf[..., n:2*n] + ghf[..., n:2*n])
+            nn = torch.tanh(gxf[..., 2*n:] + r * ghf[..., 2*n:])
+            h_new = ((1.0 - z) * nn + z * hf).to(common)
+            cache = torch.cat([r, z, nn, ghf[..., 2*n:]], dim=-1).to(common)
+            hip = False
+        ctx.save_for_backward(h_, cache)
+        ctx.hip = hip
+        ctx.in_dtypes = (gx.dtype, gh.dtype, h.dtype)
         return h_new
 
     @staticmethod
     def backward(ctx, dh_new: Tensor):
         h, cache = ctx.saved_tensors
+        gxd, ghd, hd = ctx.in_dtypes
         dh_new = dh_new.contiguous()
         if ctx.hip:
             from . import ext
-            dgx, dgh, dh = ext.require().gru_gates_bwd(dh_new, h, cache)
-            return dgx, dgh, dh
+            dgx, dgh, dh = ext.require().gru_gates_bwd(dh_new.to(h.dtype), h, cache)
+            return dgx.to(gxd), dgh.to(ghd), dh.to(hd)
         n = h.shape[-1]
-        r, z, nn, ghn = cache[..., :n], cache[..., n:2*n], cache[..., 2*n:3*n], cache[..., 3*n:]
-        dz = dh_new * (h - nn) * z * (1.0 - z)
-        dnn_pre = dh_new * (1.0 - z) * (1.0 - nn * nn)
+        cf = _up(cache)
+        r, z, nn, ghn = cf[..., :n], cf[..., n:2*n], cf[..., 2*n:3*n], cf[..., 3*n:]
+        g = _up(dh_new)
+        dz = g * (_up(h) - nn) * z * (1.0 - z)
+        dnn_pre = g * (1.0 - z) * (1.0 - nn * nn)
         dr = dnn_pre * ghn * r * (1.0 - r)
         dgx = torch.cat([dr, dz, dnn_pre], dim=-1)
         dgh = torch.cat([dr, dz, dnn_pre * r], dim=-1)
-        dh = dh_new * z
-        return dgx, dgh, dh
+        dh = g * z
+        return dgx.to(gxd), dgh.to(ghd), dh.to(hd)
 
 
 @torch.no_grad()
@@ -196,14 +214,15 @@ def fused_gru_cell(
     w_x: Tensor, w_h: Tensor,
     b_x: Optional[Tensor] = None, b_h: Optional[Tensor] = None,
 ) -> Tensor:
-    """Decode-path GRU step: single fused launch (GEMMs + gates) on HIP,
-    eager composition elsewhere.  Inference only (no autograd)."""
+    """Decode-path GRU step: single fused launch (gate GEMVs + nonlinearity)
+    on HIP, eager composition elsewhere.  Inference only (no autograd)."""
     if use_hip(x):
         from . import ext
+        common = torch.promote_types(x.dtype, h.dtype)
         return ext.require().fused_gru_cell(
-            x.contiguous(), h.contiguous(),
-            w_x.contiguous(), w_h.contiguous(),
-            b_x.contiguous() if b_x is not None else None,
-            b_h.contiguous() if b_h is not None else None,
+            x.to(common).contiguous(), h.to(common).contiguous(),
+            w_x.float().contiguous(), w_h.float().contiguous(),
+            b_x.float().contiguous() if b_x is not None else None,
+            b_h.float().contiguous() if b_h is not None else None,
         )
     return gru_cell(x, h, w_x, w_h, b_x, b_h)
