@@ -101,3 +101,21 @@ def test_decode_fused_matches_unfused():
     fused = greedy_decode(m, x, xm, maxlen=12, fused=True)
     unfused = greedy_decode(m, x, xm, maxlen=12, fused=False)
     assert fused == unfused
+
+
+def test_graph_captured_decode_matches_eager_loop():
+    """hipGraph-captured beam step must produce the same hypotheses as the
+    non-captured loop (fp32; argmax-stable)."""
+    from wap_amd import ops
+    from wap_amd.models import WAPModel
+    from wap_amd.runtime import beam_decode
+    ops.set_backend("auto")
+    torch.manual_seed(4)
+    dev = torch.device("cuda", 0)
+    m = WAPModel(n_tokens=40, encoder="vgg", embed_dim=32, hidden_dim=32,
+                 attention_dim=32, coverage_ch=16, coverage_kernel=5).to(dev).eval()
+    x = torch.rand(2, 1, 48, 96, device=dev)
+    xm = torch.ones(2, 48, 96, device=dev)
+    plain = beam_decode(m, x, xm, beam=5, maxlen=16, use_graph=False)
+    graphed = beam_decode(m, x, xm, beam=5, maxlen=16, use_graph=True)
+    assert plain == graphed

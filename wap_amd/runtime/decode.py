@@ -5,15 +5,19 @@ for the k live hypotheses (batched), top-k over k*K expansions, hypotheses
 emitting <eos>=0 die; final ranking by length-normalized score (score/len);
 greedy = the k=1 path.
 
-MI355X-first design: the whole beam step is tensorized with STATIC shapes
-(always B*k rows; dead hypotheses are padded, never dropped) so the per-token
-step is hipGraph-capturable (`CapturedBeamStep`): on ROCm the graph replay
-removes the per-launch host overhead that dominates this latency-bound loop.
+MI355X-first design: the per-token step is a single sync-free tensor
+function over STATIC [B*k, ...] state buffers (dead hypotheses are padded,
+never dropped), updated in place — so the whole step (two fused GRU cells,
+coverage conv, fused attention, readout, top-k, state reorder) is
+hipGraph-captured once and replayed per token (``use_graph=True``): replay
+removes the per-launch host overhead that dominates this latency-bound
+loop (MI355X_MICROARCH.md row graph-replay-floor).  The host only syncs on
+the liveness flag every ``sync_every`` tokens.
 """
 
 from __future__ import annotations
 
-from typing import List, Optional, Tuple
+from typing import Dict, List, Optional
 
 import torch
 from torch import Tensor
@@ -29,6 +33,41 @@ def _expand_for_beams(t: Tensor, k: int) -> Tensor:
     return t.repeat_interleave(k, dim=0)
 
 
+@torch.no_grad()
+def _beam_step_(model, S: Dict[str, Tensor], b: int, k: int, fused: bool) -> None:
+    """One beam expansion; updates every state buffer of S IN PLACE with no
+    host synchronization and static shapes (hipGraph-capturable)."""
+    m = model.parser
+    hp, wp = S["coverage"].shape[1], S["coverage"].shape[2]
+    logits, s_new, _, alpha = m.step(
+        S["emb"], S["s"], S["a_flat"], S["ua"], S["coverage"], S["mask"],
+        fused=fused,
+    )
+    logprob = torch.log_softmax(logits.float(), dim=-1).view(b, k, -1)
+    K = logprob.shape[-1]
+    # Dead hypotheses may only emit <eos> at zero cost (score frozen).
+    dead = ~S["alive"]
+    frozen = torch.full_like(logprob, NEG_INF)
+    frozen[:, :, EOS_ID] = 0.0
+    logprob = torch.where(dead.unsqueeze(-1), frozen, logprob)
+    total = S["scores"].unsqueeze(-1) + logprob               # [B,k,K]
+    top_scores, flat_idx = total.view(b, k * K).topk(k, dim=1)
+    parent = flat_idx // K                                    # [B,k]
+    token = flat_idx % K                                      # [B,k]
+
+    rows = (parent + S["brow"]).view(-1)                      # [B*k]
+    S["s"].copy_(s_new[rows])
+    new_cov = S["coverage"] + alpha.view(b * k, hp, wp).to(S["coverage"].dtype)
+    S["coverage"].copy_(new_cov[rows])
+    parent_alive = torch.gather(S["alive"], 1, parent)
+    S["lengths"].copy_(torch.gather(S["lengths"], 1, parent) + parent_alive.long())
+    S["alive"].copy_(parent_alive & (token != EOS_ID))
+    S["scores"].copy_(top_scores)
+    S["token"].copy_(token)
+    S["parent"].copy_(parent)
+    S["emb"].copy_(model.parser.embedding(token.view(-1)).to(S["emb"].dtype))
+
+
 class BeamState:
     """Static-shape beam state: all tensors keep [B*k, ...] layout."""
 
@@ -39,68 +78,71 @@ class BeamState:
         b, d, hp, wp = a.shape
         self.b, self.hp, self.wp = b, hp, wp
         a_flat, ua = model.parser.attention.precompute(a)
-        self.a_flat = _expand_for_beams(a_flat, beam)
-        self.ua = _expand_for_beams(ua, beam)
-        self.mask = _expand_for_beams(mask, beam)
-        mask_flat = self.mask.flatten(1)
-        self.s = model.parser.init_state(self.a_flat, mask_flat)
-        self.coverage = torch.zeros(b * beam, hp, wp, device=a.device, dtype=a.dtype)
-        self.emb_prev = torch.zeros(b * beam, model.parser.embedding.embedding_dim,
-                                    device=a.device, dtype=a.dtype)
-        self.scores = torch.full((b, beam), NEG_INF, device=a.device)
-        self.scores[:, 0] = 0.0
-        self.alive = torch.ones(b, beam, dtype=torch.bool, device=a.device)
-        self.lengths = torch.zeros(b, beam, dtype=torch.long, device=a.device)
-        self.tokens: List[Tensor] = []     # list of [B, k] per step
-        self.parents: List[Tensor] = []    # list of [B, k] back-pointers
+        dev = a.device
+        mask_bk = _expand_for_beams(mask, beam)
+        a_flat_bk = _expand_for_beams(a_flat, beam).contiguous()
+        s0 = model.parser.init_state(a_flat_bk, mask_bk.flatten(1))
+        self.S: Dict[str, Tensor] = {
+            "a_flat": a_flat_bk,
+            "ua": _expand_for_beams(ua, beam).contiguous(),
+            "mask": mask_bk.contiguous(),
+            "s": s0.contiguous(),
+            "coverage": torch.zeros(b * beam, hp, wp, device=dev, dtype=a.dtype),
+            "emb": torch.zeros(b * beam, model.parser.embedding.embedding_dim,
+                               device=dev, dtype=s0.dtype),
+            "scores": torch.full((b, beam), NEG_INF, device=dev),
+            "alive": torch.ones(b, beam, dtype=torch.bool, device=dev),
+            "lengths": torch.zeros(b, beam, dtype=torch.long, device=dev),
+            "token": torch.zeros(b, beam, dtype=torch.long, device=dev),
+            "parent": torch.zeros(b, beam, dtype=torch.long, device=dev),
+            "brow": (torch.arange(b, device=dev) * beam).unsqueeze(1),
+        }
+        self.S["scores"][:, 0] = 0.0
+        self.tokens: List[Tensor] = []     # [B, k] per step
+        self.parents: List[Tensor] = []    # [B, k] back-pointers
+        self.graph = None
 
-    def step(self, fused: bool = True) -> bool:
-        """One expansion step.  Returns True when every hypothesis is dead."""
-        m = self.model.parser
-        b, k = self.b, self.k
-        logits, s_new, _, alpha = m.step(
-            self.emb_prev, self.s, self.a_flat, self.ua, self.coverage,
-            self.mask, fused=fused,
-        )
-        logprob = torch.log_softmax(logits.float(), dim=-1).view(b, k, -1)
-        K = logprob.shape[-1]
-        # Dead hypotheses may only emit <eos> at zero cost (score frozen).
-        dead = ~self.alive
-        if dead.any():
-            frozen = torch.full_like(logprob, NEG_INF)
-            frozen[:, :, EOS_ID] = 0.0
-            logprob = torch.where(dead.unsqueeze(-1), frozen, logprob)
-        total = self.scores.unsqueeze(-1) + logprob          # [B,k,K]
-        top_scores, flat_idx = total.view(b, k * K).topk(k, dim=1)
-        parent = flat_idx // K                               # [B,k]
-        token = flat_idx % K                                 # [B,k]
+    def step(self, fused: bool = True) -> None:
+        _beam_step_(self.model, self.S, self.b, self.k, fused)
+        self.record()
 
-        # Gather/reorder states by parent beam.
-        gather_rows = (parent + torch.arange(b, device=parent.device)
-                       .unsqueeze(1) * k).view(-1)           # [B*k]
-        self.s = s_new[gather_rows]
-        new_cov = self.coverage + alpha.view(b * k, self.hp, self.wp)
-        self.coverage = new_cov[gather_rows]
-        parent_alive = torch.gather(self.alive, 1, parent)
-        self.lengths = torch.gather(self.lengths, 1, parent) + parent_alive.long()
-        self.alive = parent_alive & (token != EOS_ID)
-        self.scores = top_scores
-        self.tokens.append(token)
-        self.parents.append(parent)
-        emb = self.model.parser.embedding(token.view(-1))
-        # Dead rows keep the zero embedding (value irrelevant: frozen).
-        self.emb_prev = emb.to(self.emb_prev.dtype)
-        return bool((~self.alive).all().item())
+    def record(self) -> None:
+        self.tokens.append(self.S["token"].clone())
+        self.parents.append(self.S["parent"].clone())
+
+    # ---- hipGraph capture -------------------------------------------------
+    def capture(self, fused: bool = True, autocast_bf16: bool = False) -> None:
+        """Capture one beam step into a hipGraph (CUDAGraph on PyTorch-ROCm);
+        subsequent ``replay()`` calls re-run it on the static buffers."""
+        torch.cuda.synchronize()
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):  # warm up allocator/MIOpen off-graph
+            for _ in range(2):
+                _beam_step_(self.model, self.S, self.b, self.k, fused)
+        torch.cuda.current_stream().wait_stream(side)
+        torch.cuda.synchronize()
+        g = torch.cuda.CUDAGraph()
+        ctx = torch.autocast("cuda", dtype=torch.bfloat16, enabled=autocast_bf16)
+        with ctx, torch.cuda.graph(g):
+            _beam_step_(self.model, self.S, self.b, self.k, fused)
+        self.graph = g
+
+    def replay(self) -> None:
+        self.graph.replay()
+        self.record()
+
+    def all_dead(self) -> bool:
+        return bool((~self.S["alive"]).all().item())
 
     def extract(self) -> List[List[int]]:
         """Backtrace the best (length-normalized) hypothesis per image."""
         b, k = self.b, self.k
-        norm = self.scores / self.lengths.clamp_min(1).float()
-        best = norm.argmax(dim=1)                            # [B]
+        norm = self.S["scores"] / self.S["lengths"].clamp_min(1).float()
+        best = norm.argmax(dim=1).cpu()
         out: List[List[int]] = []
         tokens = [t.cpu() for t in self.tokens]
         parents = [p.cpu() for p in self.parents]
-        best = best.cpu()
         for i in range(b):
             beam_idx = int(best[i])
             seq: List[int] = []
@@ -122,18 +164,39 @@ def beam_decode(
     beam: int = 10,
     maxlen: int = 200,
     fused: bool = True,
+    use_graph: Optional[bool] = None,
+    sync_every: int = 8,
 ) -> List[List[int]]:
     """Beam-search decode a batch -> token id lists (no eos)."""
     model.eval()
     state = BeamState(model, x, x_mask, beam)
-    for _ in range(maxlen):
-        if state.step(fused=fused):
-            break
+    if use_graph is None:
+        use_graph = x.is_cuda
+    if use_graph and x.is_cuda:
+        # Graph capture mutates state (warmup steps run on the real buffers):
+        # snapshot/restore the mutable state around capture.
+        mutable = ["s", "coverage", "emb", "scores", "alive", "lengths"]
+        saved = {n: state.S[n].clone() for n in mutable}
+        autocast_on = torch.is_autocast_enabled("cuda")
+        state.capture(fused=fused, autocast_bf16=autocast_on)
+        for n in mutable:
+            state.S[n].copy_(saved[n])
+        for t in range(maxlen):
+            state.replay()
+            if (t + 1) % sync_every == 0 and state.all_dead():
+                break
+    else:
+        for t in range(maxlen):
+            state.step(fused=fused)
+            if state.all_dead():
+                break
     return state.extract()
 
 
 @torch.no_grad()
 def greedy_decode(
-    model, x: Tensor, x_mask: Tensor, maxlen: int = 200, fused: bool = True
+    model, x: Tensor, x_mask: Tensor, maxlen: int = 200, fused: bool = True,
+    use_graph: Optional[bool] = None,
 ) -> List[List[int]]:
-    return beam_decode(model, x, x_mask, beam=1, maxlen=maxlen, fused=fused)
+    return beam_decode(model, x, x_mask, beam=1, maxlen=maxlen, fused=fused,
+                       use_graph=use_graph)
