@@ -113,7 +113,9 @@ __global__ void attn_score_fwd_kernel(
     for (int k = lane; k < A; k += WAVE) {
       const float t = tanhf(qs[k] + ldf(ua + off + k) + ldf(uf + off + k));
       stf(tcache + off + k, t);
-      acc += ns[k] * t;
+      // accumulate the ROUNDED value (what the cache holds) so forward and
+      // backward see the same t, and so bf16 matches the eager oracle
+      acc += ns[k] * ldf(tcache + off + k);
     }
     acc = wave_sum(acc);
     if (lane == 0) e[(size_t)b * L + l] = acc;
