@@ -45,6 +45,15 @@ class CoverageAttention(nn.Module):
         ua = self.U_a(a_flat)                               # [B,L,A]
         return a_flat, ua
 
+    def merged_coverage_weight(self) -> Tensor:
+        """U_f o conv_Q collapses to ONE 11x11 conv: conv_Q has 1 input
+        channel, so U_f(conv_Q(beta)) == conv2d(beta, U_f.weight @ Q).
+        Tiny [A,q]@[q,KK] GEMM per step; removes the [B*L,q]x[q,A]
+        projection GEMM and the F tensor from the per-step path."""
+        q_ch, _, kh, kw = self.conv_Q.weight.shape
+        w = self.U_f.weight @ self.conv_Q.weight.view(q_ch, kh * kw)
+        return w.view(self.attention_dim, 1, kh, kw)
+
     def forward(
         self,
         s_hat: Tensor,      # [B, n] pre-attention decoder state
@@ -55,9 +64,12 @@ class CoverageAttention(nn.Module):
     ) -> Tuple[Tensor, Tensor]:
         """Returns (context[B,D], alpha[B,L])."""
         b, hp, wp = coverage.shape
-        f = self.conv_Q(coverage.unsqueeze(1))              # [B,q,H',W']
-        f_flat = f.flatten(2).transpose(1, 2)               # [B,L,q]
-        uf = self.U_f(f_flat)                               # [B,L,A]
+        w = self.merged_coverage_weight()
+        kh = w.shape[2]
+        uf = torch.nn.functional.conv2d(
+            coverage.unsqueeze(1), w, padding=kh // 2
+        )                                                   # [B,A,H',W']
+        uf = uf.flatten(2).transpose(1, 2) + self.U_f.bias  # [B,L,A]
         q = self.W_a(s_hat)                                 # [B,A]
         alpha, ctx = ops.coverage_attention(
             q, ua, uf, self.nu, a_flat, mask.flatten(1)
