@@ -66,10 +66,11 @@ class CoverageAttention(nn.Module):
         b, hp, wp = coverage.shape
         w = self.merged_coverage_weight()
         kh = w.shape[2]
-        uf = torch.nn.functional.conv2d(
-            coverage.unsqueeze(1), w, padding=kh // 2
-        )                                                   # [B,A,H',W']
-        uf = uf.flatten(2).transpose(1, 2) + self.U_f.bias  # [B,L,A]
+        # 1-channel conv via unfold+GEMM (MIOpen naive-kernel bypass, see
+        # encoder.conv_1ch) — this runs once per decode timestep.
+        from .encoder import conv_1ch
+        uf = conv_1ch(coverage.unsqueeze(1), w, padding=kh // 2)  # [B,A,H',W']
+        uf = uf.flatten(2).transpose(1, 2) + self.U_f.bias        # [B,L,A]
         q = self.W_a(s_hat)                                 # [B,A]
         alpha, ctx = ops.coverage_attention(
             q, ua, uf, self.nu, a_flat, mask.flatten(1)

@@ -25,6 +25,27 @@ from torch import Tensor, nn
 import torch.nn.functional as F
 
 
+def conv_1ch(x: Tensor, weight: Tensor, bias=None, stride: int = 1,
+             padding: int = 0) -> Tensor:
+    """Conv2d for 1-input-channel tensors via unfold + rocBLAS GEMM.
+
+    MIOpen (ROCm 7.2) falls back to its NAIVE nchw kernels for 1-channel
+    bf16 convs — measured 74% of the whole WAP train step (fwd+dgrad+wrw;
+    profiles/r01_baseline_kernel_stats.md).  unfold+matmul routes the same
+    math through hipBLASLt instead.  x[B,1,H,W], weight[C,1,kh,kw].
+    """
+    b = x.shape[0]
+    c, _, kh, kw = weight.shape
+    ho = (x.shape[2] + 2 * padding - kh) // stride + 1
+    wo = (x.shape[3] + 2 * padding - kw) // stride + 1
+    cols = torch.nn.functional.unfold(x, (kh, kw), padding=padding,
+                                      stride=stride)          # [B,kh*kw,L]
+    out = weight.view(c, kh * kw) @ cols                      # [B,C,L]
+    if bias is not None:
+        out = out + bias.view(1, c, 1)
+    return out.view(b, c, ho, wo)
+
+
 def downsample_mask(x_mask: Tensor, stride: int) -> Tensor:
     """[B,H,W] pixel mask -> [B,H',W'] feature-grid mask (max-pool)."""
     m = x_mask.unsqueeze(1)
@@ -89,7 +110,7 @@ class DenseNetEncoder(nn.Module):
         self.out_channels = ch  # 684 with defaults
 
     def forward(self, x: Tensor, x_mask: Tensor) -> Tuple[Tensor, Tensor]:
-        h = self.stem(x)
+        h = conv_1ch(x, self.stem.weight, stride=2, padding=3)
         h = F.relu(self.stem_bn(h))
         h = F.max_pool2d(h, 2, ceil_mode=True)   # stride 4 after stem+pool
         a = self.blocks(h)
