@@ -81,9 +81,11 @@ def validate(model, buckets, vocab, cfg, device, autocast_ctx) -> float:
     with torch.no_grad():
         for batch in loader:
             with autocast_ctx():
+                # use_graph=False: validation buckets have many distinct
+                # shapes, per-shape capture would dominate
                 hyps = greedy_decode(
                     model, batch["x"], batch["x_mask"],
-                    maxlen=cfg.decode_maxlen, fused=True,
+                    maxlen=cfg.decode_maxlen, fused=True, use_graph=False,
                 )
             refs = batch["y"]
             masks = batch["y_mask"]

@@ -102,6 +102,19 @@ class BeamState:
         self.parents: List[Tensor] = []    # [B, k] back-pointers
         self.graph = None
 
+    def reset_from(self, other: "BeamState") -> None:
+        """Reuse this state's (captured) buffers for a new decode problem of
+        the same shape: copy the per-problem tensors in, reset the mutable
+        state.  The hipGraph stays valid (same storage)."""
+        S, O = self.S, other.S
+        for n in ("a_flat", "ua", "mask", "s", "coverage", "emb"):
+            S[n].copy_(O[n])
+        S["scores"].fill_(NEG_INF)
+        S["scores"][:, 0] = 0.0
+        S["alive"].fill_(True)
+        S["lengths"].zero_()
+        self.tokens, self.parents = [], []
+
     def step(self, fused: bool = True) -> None:
         _beam_step_(self.model, self.S, self.b, self.k, fused)
         self.record()
@@ -156,6 +169,20 @@ class BeamState:
         return out
 
 
+# Captured-graph cache: capture costs ~100 ms, so graphs + their static
+# buffers persist across beam_decode calls, keyed by problem shape.  Replay
+# reads weights through the live parameter storage, so training in between
+# does not invalidate a graph.  Bounded LRU.
+_GRAPH_CACHE: Dict[tuple, BeamState] = {}
+_GRAPH_CACHE_CAP = 16
+
+
+def _graph_key(model, state: BeamState, fused: bool, autocast_on: bool):
+    return (id(model), state.b, state.k, state.hp, state.wp,
+            state.S["a_flat"].shape[-1], str(state.S["a_flat"].dtype),
+            fused, autocast_on)
+
+
 @torch.no_grad()
 def beam_decode(
     model,
@@ -173,14 +200,19 @@ def beam_decode(
     if use_graph is None:
         use_graph = x.is_cuda
     if use_graph and x.is_cuda:
-        # Graph capture mutates state (warmup steps run on the real buffers):
-        # snapshot/restore the mutable state around capture.
-        mutable = ["s", "coverage", "emb", "scores", "alive", "lengths"]
-        saved = {n: state.S[n].clone() for n in mutable}
         autocast_on = torch.is_autocast_enabled("cuda")
-        state.capture(fused=fused, autocast_bf16=autocast_on)
-        for n in mutable:
-            state.S[n].copy_(saved[n])
+        key = _graph_key(model, state, fused, autocast_on)
+        cached = _GRAPH_CACHE.get(key)
+        if cached is None:
+            state.capture(fused=fused, autocast_bf16=autocast_on)
+            # capture mutated the state: re-init from a fresh encode
+            state.reset_from(BeamState(model, x, x_mask, beam))
+            if len(_GRAPH_CACHE) >= _GRAPH_CACHE_CAP:
+                _GRAPH_CACHE.pop(next(iter(_GRAPH_CACHE)))
+            _GRAPH_CACHE[key] = state
+        else:
+            cached.reset_from(state)
+            state = cached
         for t in range(maxlen):
             state.replay()
             if (t + 1) % sync_every == 0 and state.all_dead():
