@@ -166,3 +166,21 @@ def test_all_masked_row_is_safe():
     alpha, ctx = masked_softmax_context(e, a, mask)
     assert not torch.isnan(alpha).any() and not torch.isnan(ctx).any()
     assert (alpha[1] == 0).all() and (ctx[1] == 0).all()
+
+
+@pytest.mark.parametrize("stride,pad,k,H,W", [(1, 5, 11, 8, 32), (2, 3, 7, 37, 53), (1, 1, 3, 16, 16)])
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_im2col_col2im_1ch(stride, pad, k, H, W, dtype):
+    from wap_amd.ops import unfold_1ch
+    dev = "cuda"
+    x = torch.randn(3, 1, H, W, device=dev, dtype=dtype, requires_grad=True)
+    cols = unfold_1ch(x, k, k, stride, pad)
+    ref = torch.nn.functional.unfold(x.float(), (k, k), padding=pad, stride=stride)
+    assert cols.shape == ref.shape
+    assert _rel_err(cols, ref) < (1e-6 if dtype == torch.float32 else 1e-2)
+    g = torch.randn_like(cols, dtype=torch.float32).to(dtype)
+    (dx,) = torch.autograd.grad(cols, x, g)
+    x2 = x.detach().float().requires_grad_()
+    ref2 = torch.nn.functional.unfold(x2, (k, k), padding=pad, stride=stride)
+    (dx_ref,) = torch.autograd.grad(ref2, x2, g.float())
+    assert _rel_err(dx, dx_ref) < (1e-6 if dtype == torch.float32 else 1e-2)

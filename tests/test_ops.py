@@ -109,3 +109,21 @@ def test_gru_cell_backward_matches_autograd():
     assert torch.allclose(o1, o2, atol=1e-12)
     assert torch.allclose(gx1, gx2, atol=1e-10)
     assert torch.allclose(gh1, gh2, atol=1e-10)
+
+
+def test_conv_1ch_matches_conv2d():
+    """unfold+GEMM conv (MIOpen naive-fallback bypass) == F.conv2d."""
+    from wap_amd.models.encoder import conv_1ch
+    for stride, pad, k in [(1, 5, 11), (2, 3, 7), (1, 1, 3)]:
+        x = torch.randn(2, 1, 20, 26, dtype=torch.double, requires_grad=True)
+        w = torch.randn(6, 1, k, k, dtype=torch.double, requires_grad=True)
+        bias = torch.randn(6, dtype=torch.double, requires_grad=True)
+        out = conv_1ch(x, w, bias, stride=stride, padding=pad)
+        ref = torch.nn.functional.conv2d(x, w, bias, stride=stride, padding=pad)
+        assert torch.allclose(out, ref, atol=1e-10), (stride, pad, k)
+        g = torch.randn_like(out)
+        gx1, gw1, gb1 = torch.autograd.grad(out, (x, w, bias), g, retain_graph=True)
+        gx2, gw2, gb2 = torch.autograd.grad(ref, (x, w, bias), g)
+        assert torch.allclose(gx1, gx2, atol=1e-10)
+        assert torch.allclose(gw1, gw2, atol=1e-10)
+        assert torch.allclose(gb1, gb2, atol=1e-10)

@@ -124,26 +124,36 @@ __global__ void attn_score_fwd_kernel(
 
 // backward: g[b,l,k] = de[b,l] * nu[k] * (1 - t^2)
 //   dua = g (duf identical), dq[b,k] = sum_l g, dnu[k] = sum_{b,l} t * de
+// Grid (l-chunks, B): per-thread partial dq/dnu accumulate over the
+// block's l's in registers, one atomicAdd per (block, k) at the end.
+#define ATTN_BWD_LCH 16
 template <typename T>
 __global__ void attn_score_bwd_kernel(
     const float* __restrict__ de, const T* __restrict__ t,
-    const float* __restrict__ nu, T* __restrict__ dq, T* __restrict__ dua,
-    float* __restrict__ dnu, int L, int A) {
-  const int b = blockIdx.x;
-  for (int k = threadIdx.x; k < A; k += blockDim.x) {
-    const float nk = nu[k];
-    float accq = 0.f, accn = 0.f;
-    for (int l = 0; l < L; ++l) {
-      const size_t off = ((size_t)b * L + l) * A + k;
-      const float tv = ldf(t + off);
-      const float d = de[(size_t)b * L + l];
-      const float g = d * nk * (1.f - tv * tv);
-      stf(dua + off, g);
-      accq += g;
-      accn += tv * d;
+    const float* __restrict__ nu, float* __restrict__ dq,
+    T* __restrict__ dua, float* __restrict__ dnu, int L, int A) {
+  const int b = blockIdx.y;
+  const int l0 = blockIdx.x * ATTN_BWD_LCH;
+  const int l1 = min(l0 + ATTN_BWD_LCH, L);
+  float accq[4], accn[4];  // A <= 4*blockDim.x
+#pragma unroll
+  for (int i = 0; i < 4; ++i) accq[i] = accn[i] = 0.f;
+  for (int l = l0; l < l1; ++l) {
+    const float d = de[(size_t)b * L + l];
+    const size_t row = ((size_t)b * L + l) * A;
+    int i = 0;
+    for (int k = threadIdx.x; k < A; k += blockDim.x, ++i) {
+      const float tv = ldf(t + row + k);
+      const float g = d * nu[k] * (1.f - tv * tv);
+      stf(dua + row + k, g);
+      accq[i] += g;
+      accn[i] += tv * d;
     }
-    stf(dq + (size_t)b * A + k, accq);
-    atomicAdd(dnu + k, accn);
+  }
+  int i = 0;
+  for (int k = threadIdx.x; k < A; k += blockDim.x, ++i) {
+    atomicAdd(dq + (size_t)b * A + k, accq[i]);
+    atomicAdd(dnu + k, accn[i]);
   }
 }
 
@@ -194,47 +204,59 @@ __global__ void msc_fwd_kernel(
   }
 }
 
-// backward:
-//   t_l  = g_alpha_l + g_ctx . a_l          (staged into de[] as scratch)
-//   de_l = alpha_l * (t_l - sum_m alpha_m t_m)
-//   da[l,d] = alpha_l * g_ctx[d]
+// backward (two launches, grid (l-chunks, B) each):
+//   pass A: t_l = g_alpha_l + g_ctx . a_l  -> de[] scratch;
+//           partial sum alpha_l*t_l -> atomicAdd sscr[b];
+//           da[l,d] = alpha_l * g_ctx[d]
+//   pass B: de_l = alpha_l * (t_l - sscr[b])
+#define MSC_BWD_LCH 8
 template <typename T>
-__global__ void msc_bwd_kernel(
+__global__ void msc_bwd_a_kernel(
     const float* __restrict__ g_alpha, const float* __restrict__ g_ctx,
     const float* __restrict__ alpha, const T* __restrict__ a,
-    float* __restrict__ de, T* __restrict__ da, int L, int D) {
+    float* __restrict__ de, T* __restrict__ da, float* __restrict__ sscr,
+    int L, int D) {
   extern __shared__ float smem[];
   float* red = smem;           // [16]
   float* gc = smem + 16;       // [D]
-  const int b = blockIdx.x;
+  const int b = blockIdx.y;
   g_alpha += (size_t)b * L;
   g_ctx += (size_t)b * D;
   alpha += (size_t)b * L;
   a += (size_t)b * L * D;
   de += (size_t)b * L;
   da += (size_t)b * L * D;
-
   for (int d = threadIdx.x; d < D; d += blockDim.x) gc[d] = g_ctx[d];
   __syncthreads();
-
   const int lane = threadIdx.x & (WAVE - 1), w = threadIdx.x >> 6;
   const int nw = blockDim.x / WAVE;
-  for (int l = w; l < L; l += nw) {
-    float acc = 0.f;
-    for (int d = lane; d < D; d += WAVE) acc += gc[d] * ldf(a + (size_t)l * D + d);
-    acc = wave_sum(acc);
-    if (lane == 0) de[l] = g_alpha[l] + acc;  // de used as t-scratch
-  }
-  __syncthreads();
+  const int l0 = blockIdx.x * MSC_BWD_LCH;
+  const int l1 = min(l0 + MSC_BWD_LCH, L);
   float sp = 0.f;
-  for (int l = threadIdx.x; l < L; l += blockDim.x) sp += alpha[l] * de[l];
-  sp = block_sum<256>(sp, red);
-  for (int l = threadIdx.x; l < L; l += blockDim.x)
-    de[l] = alpha[l] * (de[l] - sp);
-  for (size_t i = threadIdx.x; i < (size_t)L * D; i += blockDim.x) {
-    const int l = i / D, d = i % D;
-    stf(da + i, alpha[l] * gc[d]);
+  for (int l = l0 + w; l < l1; l += nw) {
+    float acc = 0.f;
+    const size_t row = (size_t)l * D;
+    for (int d = lane; d < D; d += WAVE) {
+      acc += gc[d] * ldf(a + row + d);
+      stf(da + row + d, alpha[l] * gc[d]);
+    }
+    acc = wave_sum(acc);
+    acc = __shfl(acc, 0, WAVE);
+    if (lane == 0) de[l] = g_alpha[l] + acc;
+    sp += alpha[l] * (g_alpha[l] + acc) / WAVE;  // spread over lanes
   }
+  sp = block_sum<256>(sp, red);
+  if (threadIdx.x == 0) atomicAdd(sscr + b, sp);
+}
+
+__global__ void msc_bwd_b_kernel(
+    const float* __restrict__ alpha, const float* __restrict__ sscr,
+    float* __restrict__ de, int L) {
+  const int b = blockIdx.y;
+  const float s = sscr[b];
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < L;
+       i += (long)gridDim.x * blockDim.x)
+    de[(size_t)b * L + i] = alpha[(size_t)b * L + i] * (de[(size_t)b * L + i] - s);
 }
 
 // -------------------------------------------------- GRU gate nonlinearity
@@ -288,6 +310,63 @@ __global__ void gru_gates_bwd_kernel(
   stf(dgh + o3 + n + j, dz);
   stf(dgh + o3 + 2 * n + j, dnn_pre * r);
   stf(dh + i, g * z);
+}
+
+// ---------------------------------------------- batched 1-channel im2col
+// torch's unfold launches one im2col kernel PER IMAGE (measured 16
+// launches/decoder-timestep = 12% of the train step); this is one launch
+// for the whole batch.
+
+template <typename T>
+__global__ void im2col_1ch_kernel(
+    const T* __restrict__ x, T* __restrict__ cols,
+    int H, int W, int KH, int KW, int stride, int pad, int HO, int WO) {
+  const int L = HO * WO, KK = KH * KW;
+  const int b = blockIdx.y;
+  x += (size_t)b * H * W;
+  cols += (size_t)b * KK * L;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < (long)KK * L;
+       i += (long)gridDim.x * blockDim.x) {
+    const int l = i % L, j = i / L;     // cols fastest dim = l (coalesced)
+    const int yo = l / WO, xo = l % WO;
+    const int dy = j / KW, dx = j % KW;
+    const int yi = yo * stride - pad + dy;
+    const int xi = xo * stride - pad + dx;
+    float v = 0.f;
+    if (yi >= 0 && yi < H && xi >= 0 && xi < W) v = ldf(x + (size_t)yi * W + xi);
+    stf(cols + i, v);
+  }
+}
+
+// col2im (backward of im2col): dx[b,p] = sum over the <=KK kernel taps
+// that could have read pixel p.  One thread per input pixel.
+template <typename T>
+__global__ void col2im_1ch_kernel(
+    const T* __restrict__ dcols, float* __restrict__ dx,
+    int H, int W, int KH, int KW, int stride, int pad, int HO, int WO) {
+  const int L = HO * WO;
+  const int b = blockIdx.y;
+  dcols += (size_t)b * (size_t)KH * KW * L;
+  dx += (size_t)b * H * W;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < (long)H * W;
+       i += (long)gridDim.x * blockDim.x) {
+    const int yi = i / W, xi = i % W;
+    float acc = 0.f;
+    for (int dy = 0; dy < KH; ++dy) {
+      const int ys = yi + pad - dy;
+      if (ys < 0 || ys % stride) continue;
+      const int yo = ys / stride;
+      if (yo >= HO) continue;
+      for (int dx_ = 0; dx_ < KW; ++dx_) {
+        const int xs = xi + pad - dx_;
+        if (xs < 0 || xs % stride) continue;
+        const int xo = xs / stride;
+        if (xo >= WO) continue;
+        acc += ldf(dcols + (size_t)(dy * KW + dx_) * L + yo * WO + xo);
+      }
+    }
+    dx[i] = acc;
+  }
 }
 
 // ------------------------------------------ fused decode-time GRU step
@@ -401,17 +480,19 @@ std::vector<torch::Tensor> attention_score_bwd(
     torch::Tensor de, torch::Tensor t, torch::Tensor nu) {
   CHECK_GPU(de); CHECK_GPU(t); CHECK_GPU(nu);
   const int B = t.size(0), L = t.size(1), A = t.size(2);
-  auto dq = torch::empty({B, A}, t.options());
+  TORCH_CHECK(A <= 4 * 256, "A too large for attn_score_bwd accumulators");
+  auto dq = torch::zeros({B, A}, t.options().dtype(at::kFloat));
   auto dua = torch::empty_like(t);
   auto dnu = torch::zeros({A}, t.options().dtype(at::kFloat));
+  const dim3 grid((L + ATTN_BWD_LCH - 1) / ATTN_BWD_LCH, B);
   if (is_bf16(t)) {
-    hipLaunchKernelGGL(attn_score_bwd_kernel<__hip_bfloat16>, dim3(B), dim3(256),
+    hipLaunchKernelGGL(attn_score_bwd_kernel<__hip_bfloat16>, grid, dim3(256),
                        0, cur_stream(),
                        cptr<float>(de), cptr<__hip_bfloat16>(t), cptr<float>(nu),
-                       ptr<__hip_bfloat16>(dq), ptr<__hip_bfloat16>(dua),
+                       ptr<float>(dq), ptr<__hip_bfloat16>(dua),
                        ptr<float>(dnu), L, A);
   } else {
-    hipLaunchKernelGGL(attn_score_bwd_kernel<float>, dim3(B), dim3(256),
+    hipLaunchKernelGGL(attn_score_bwd_kernel<float>, grid, dim3(256),
                        0, cur_stream(),
                        cptr<float>(de), cptr<float>(t), cptr<float>(nu),
                        ptr<float>(dq), ptr<float>(dua), ptr<float>(dnu), L, A);
@@ -451,20 +532,76 @@ std::vector<torch::Tensor> masked_softmax_context_bwd(
   const int B = a.size(0), L = a.size(1), D = a.size(2);
   auto de = torch::empty({B, L}, g_alpha.options().dtype(at::kFloat));
   auto da = torch::empty_like(a);
+  auto sscr = torch::zeros({B}, g_alpha.options().dtype(at::kFloat));
   const int smem = (16 + D) * sizeof(float);
+  const dim3 grid((L + MSC_BWD_LCH - 1) / MSC_BWD_LCH, B);
   if (is_bf16(a)) {
-    hipLaunchKernelGGL(msc_bwd_kernel<__hip_bfloat16>, dim3(B), dim3(256), smem,
+    hipLaunchKernelGGL(msc_bwd_a_kernel<__hip_bfloat16>, grid, dim3(256), smem,
                        cur_stream(), cptr<float>(g_alpha), cptr<float>(g_ctx),
                        cptr<float>(alpha), cptr<__hip_bfloat16>(a),
-                       ptr<float>(de), ptr<__hip_bfloat16>(da), L, D);
+                       ptr<float>(de), ptr<__hip_bfloat16>(da),
+                       ptr<float>(sscr), L, D);
   } else {
-    hipLaunchKernelGGL(msc_bwd_kernel<float>, dim3(B), dim3(256), smem,
+    hipLaunchKernelGGL(msc_bwd_a_kernel<float>, grid, dim3(256), smem,
                        cur_stream(), cptr<float>(g_alpha), cptr<float>(g_ctx),
                        cptr<float>(alpha), cptr<float>(a),
-                       ptr<float>(de), ptr<float>(da), L, D);
+                       ptr<float>(de), ptr<float>(da), ptr<float>(sscr), L, D);
   }
   HIP_OK();
+  hipLaunchKernelGGL(msc_bwd_b_kernel, dim3((L + 255) / 256, B), dim3(256), 0,
+                     cur_stream(), cptr<float>(alpha), cptr<float>(sscr),
+                     ptr<float>(de), L);
+  HIP_OK();
   return {de, da};
+}
+
+std::vector<torch::Tensor> im2col_1ch(
+    torch::Tensor x, int64_t KH, int64_t KW, int64_t stride, int64_t pad) {
+  CHECK_GPU(x);
+  const int B = x.size(0), H = x.size(x.dim() - 2), W = x.size(x.dim() - 1);
+  const int HO = (H + 2 * pad - KH) / stride + 1;
+  const int WO = (W + 2 * pad - KW) / stride + 1;
+  const long L = (long)HO * WO;
+  auto cols = torch::empty({B, KH * KW, L}, x.options());
+  const long work = KH * KW * L;
+  const dim3 grid(std::min((work + 255) / 256, 1024L), B);
+  if (is_bf16(x)) {
+    hipLaunchKernelGGL(im2col_1ch_kernel<__hip_bfloat16>, grid, dim3(256), 0,
+                       cur_stream(), cptr<__hip_bfloat16>(x),
+                       ptr<__hip_bfloat16>(cols), H, W, (int)KH, (int)KW,
+                       (int)stride, (int)pad, HO, WO);
+  } else {
+    hipLaunchKernelGGL(im2col_1ch_kernel<float>, grid, dim3(256), 0,
+                       cur_stream(), cptr<float>(x), ptr<float>(cols),
+                       H, W, (int)KH, (int)KW, (int)stride, (int)pad, HO, WO);
+  }
+  HIP_OK();
+  return {cols};
+}
+
+torch::Tensor col2im_1ch(
+    torch::Tensor dcols, int64_t H, int64_t W, int64_t KH, int64_t KW,
+    int64_t stride, int64_t pad) {
+  CHECK_GPU(dcols);
+  const int B = dcols.size(0);
+  const int HO = (H + 2 * pad - KH) / stride + 1;
+  const int WO = (W + 2 * pad - KW) / stride + 1;
+  auto dx = torch::empty({B, H, W}, dcols.options().dtype(at::kFloat));
+  const long work = H * W;
+  const dim3 grid(std::min((work + 255) / 256, 1024L), B);
+  if (is_bf16(dcols)) {
+    hipLaunchKernelGGL(col2im_1ch_kernel<__hip_bfloat16>, grid, dim3(256), 0,
+                       cur_stream(), cptr<__hip_bfloat16>(dcols),
+                       ptr<float>(dx), (int)H, (int)W, (int)KH, (int)KW,
+                       (int)stride, (int)pad, HO, WO);
+  } else {
+    hipLaunchKernelGGL(col2im_1ch_kernel<float>, grid, dim3(256), 0,
+                       cur_stream(), cptr<float>(dcols), ptr<float>(dx),
+                       (int)H, (int)W, (int)KH, (int)KW, (int)stride, (int)pad,
+                       HO, WO);
+  }
+  HIP_OK();
+  return dx;
 }
 
 std::vector<torch::Tensor> gru_gates_fwd(
@@ -555,6 +692,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attention_score_bwd", &attention_score_bwd);
   m.def("masked_softmax_context_fwd", &masked_softmax_context_fwd);
   m.def("masked_softmax_context_bwd", &masked_softmax_context_bwd);
+  m.def("im2col_1ch", &im2col_1ch);
+  m.def("col2im_1ch", &col2im_1ch);
   m.def("gru_gates_fwd", &gru_gates_fwd);
   m.def("gru_gates_bwd", &gru_gates_bwd);
   m.def("fused_gru_cell", &fused_gru_cell);

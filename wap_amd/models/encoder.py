@@ -34,12 +34,12 @@ def conv_1ch(x: Tensor, weight: Tensor, bias=None, stride: int = 1,
     profiles/r01_baseline_kernel_stats.md).  unfold+matmul routes the same
     math through hipBLASLt instead.  x[B,1,H,W], weight[C,1,kh,kw].
     """
+    from .. import ops
     b = x.shape[0]
     c, _, kh, kw = weight.shape
     ho = (x.shape[2] + 2 * padding - kh) // stride + 1
     wo = (x.shape[3] + 2 * padding - kw) // stride + 1
-    cols = torch.nn.functional.unfold(x, (kh, kw), padding=padding,
-                                      stride=stride)          # [B,kh*kw,L]
+    cols = ops.unfold_1ch(x, kh, kw, stride, padding)         # [B,kh*kw,L]
     out = weight.view(c, kh * kw) @ cols                      # [B,C,L]
     if bias is not None:
         out = out + bias.view(1, c, 1)

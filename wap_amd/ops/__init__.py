@@ -56,10 +56,11 @@ from .functional import (  # noqa: E402
     fused_gru_cell,
     gru_cell,
     masked_softmax_context,
+    unfold_1ch,
 )
 
 __all__ = [
     "set_backend", "get_backend", "use_hip",
     "attention_score", "coverage_attention", "masked_softmax_context",
-    "gru_cell", "fused_gru_cell",
+    "gru_cell", "fused_gru_cell", "unfold_1ch",
 ]

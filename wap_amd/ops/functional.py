@@ -137,6 +137,43 @@ def coverage_attention(
 
 
 # --------------------------------------------------------------------------
+# batched 1-channel unfold (im2col) — torch's unfold launches one kernel per
+# image; the HIP op is one launch for the whole batch (fwd and bwd).
+# --------------------------------------------------------------------------
+
+class _Unfold1ch(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: Tensor, kh: int, kw: int, stride: int, pad: int) -> Tensor:
+        ctx.meta = (x.shape[-2], x.shape[-1], kh, kw, stride, pad, x.dtype)
+        if use_hip(x):
+            from . import ext
+            b = x.shape[0]
+            (cols,) = ext.require().im2col_1ch(
+                x.reshape(b, x.shape[-2], x.shape[-1]).contiguous(),
+                kh, kw, stride, pad,
+            )
+            return cols
+        return torch.nn.functional.unfold(x, (kh, kw), padding=pad, stride=stride)
+
+    @staticmethod
+    def backward(ctx, dcols: Tensor):
+        H, W, kh, kw, stride, pad, dt = ctx.meta
+        if use_hip(dcols):
+            from . import ext
+            dx = ext.require().col2im_1ch(dcols.contiguous(), H, W, kh, kw,
+                                          stride, pad)
+            return dx.unsqueeze(1).to(dt), None, None, None, None
+        dx = torch.nn.functional.fold(dcols, (H, W), (kh, kw), padding=pad,
+                                      stride=stride)
+        return dx.to(dt), None, None, None, None
+
+
+def unfold_1ch(x: Tensor, kh: int, kw: int, stride: int = 1, pad: int = 0) -> Tensor:
+    """x[B,1,H,W] -> cols[B, kh*kw, L] (batched im2col, 1 input channel)."""
+    return _Unfold1ch.apply(x, kh, kw, stride, pad)
+
+
+# --------------------------------------------------------------------------
 # GRU cell (gate order: r | z | n, PyTorch GRUCell convention)
 # --------------------------------------------------------------------------
 
