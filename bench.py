@@ -47,6 +47,9 @@ def main() -> None:
     p.add_argument("--decode-steps", type=int, default=0,
                    help="decode measurement rounds (default 1 on GPU, 0 on CPU)")
     p.add_argument("--no-decode", action="store_true")
+    p.add_argument("--graph", type=str, default="auto",
+                   choices=["auto", "on", "off"],
+                   help="hipGraph-captured train step")
     args = p.parse_args()
 
     ops.set_backend(args.kernels)
@@ -77,7 +80,25 @@ def main() -> None:
     y = torch.from_numpy(y_np).to(device)
     y_mask = torch.ones(args.batch, args.seq_len, device=device)
 
-    def train_step() -> float:
+    batch = {"x": x, "x_mask": x_mask, "y": y, "y_mask": y_mask}
+    gstep = None
+    if on_gpu and args.graph != "off":
+        try:
+            from wap_amd.runtime.graph_train import GraphedTrainStep
+            gstep = GraphedTrainStep(model, ddp, opt, clip=100.0,
+                                     autocast_bf16=use_bf16)
+            gstep.initialize(batch)
+        except Exception as e:  # pragma: no cover - graph capture issues
+            if args.graph == "on":
+                raise
+            import sys
+            print(f"graph capture unavailable, eager fallback: {e!r}",
+                  file=sys.stderr)
+            gstep = None
+
+    def train_step():
+        if gstep is not None:
+            return gstep.step()
         ddp.zero_grad()
         with autocast():
             out = model(x, x_mask, y, y_mask)
@@ -156,6 +177,7 @@ def main() -> None:
                 "image": f"{args.height}x{args.width}",
                 "parallelism": f"dp{world}",
                 "kernels": ops.get_backend(),
+                "graph": gstep is not None,
             },
             "extra": extra,
         }))

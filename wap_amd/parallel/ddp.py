@@ -50,6 +50,10 @@ class BucketedDDP:
         self.model = model
         self.group = process_group
         self.world_size = dist.get_world_size(process_group) if _dist_ready() else 1
+        # When False, backward hooks do NOT launch all-reduces (used by the
+        # graphed train step, which all-reduces the flat buffers itself
+        # between graph replays).
+        self.sync_enabled = True
         self.buckets: List[Bucket] = []
         self._param_bucket: Dict[int, Bucket] = {}
         self._hooks = []
@@ -102,6 +106,8 @@ class BucketedDDP:
             b.handle = None
 
     def _on_grad_ready(self, p: nn.Parameter) -> None:
+        if not self.sync_enabled:
+            return
         b = self._param_bucket[id(p)]
         b.pending -= 1
         if b.pending == 0:

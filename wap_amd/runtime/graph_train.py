@@ -1,0 +1,113 @@
+"""hipGraph-captured training step.
+
+The WAP train step launches ~6-7k kernels (decoder unroll: T timesteps x
+~20 small kernels, forward and backward) — measured launch/gap overhead is
+a double-digit share of the step on MI355X.  For fixed-shape steps (bench;
+padded buckets) the whole step is captured once and replayed:
+
+  graph A: grad-buffer zero -> autocast forward -> backward
+  (world>1: eager bucketed RCCL all-reduce on the flat grad buffers)
+  graph B: grad clip + optimizer step
+
+The all-reduce stays OUTSIDE the graphs: capturing RCCL collectives into
+hipGraphs is not load-bearing here (grad volume ~60 MB, sub-ms over xGMI)
+and keeping it eager avoids capture-mode collective hazards.  Gradients
+already live in the flat per-bucket buffers of BucketedDDP, so the
+between-graph all-reduce is one call per bucket, no packing.
+"""
+
+from __future__ import annotations
+
+from typing import Callable, Dict, Optional
+
+import torch
+import torch.distributed as dist
+
+
+class GraphedTrainStep:
+    """Capture-and-replay train step over static input buffers.
+
+    Usage::
+
+        gstep = GraphedTrainStep(model, ddp, opt, clip=100.0,
+                                 autocast_bf16=True)
+        gstep.initialize(batch)      # 2 eager warmups + capture
+        loss_t = gstep.step(batch)   # replay (copies batch in)
+    """
+
+    def __init__(self, model, ddp, optimizer, clip: float = 100.0,
+                 autocast_bf16: bool = True):
+        self.model = model
+        self.ddp = ddp
+        self.opt = optimizer
+        self.clip = clip
+        self.autocast_bf16 = autocast_bf16
+        self.buf: Dict[str, torch.Tensor] = {}
+        self.loss: Optional[torch.Tensor] = None
+        self.g_fwdbwd: Optional[torch.cuda.CUDAGraph] = None
+        self.g_opt: Optional[torch.cuda.CUDAGraph] = None
+
+    def _autocast(self):
+        return torch.autocast("cuda", dtype=torch.bfloat16,
+                              enabled=self.autocast_bf16)
+
+    def _fwd_bwd(self):
+        for b in self.ddp.buckets:
+            b.buffer.zero_()
+        with self._autocast():
+            out = self.model(self.buf["x"], self.buf["x_mask"],
+                             self.buf["y"], self.buf["y_mask"])
+        out["loss"].backward()
+        return out["loss"]
+
+    def _clip_opt(self):
+        if self.clip > 0:
+            torch.nn.utils.clip_grad_norm_(self.model.parameters(), self.clip)
+        self.opt.step()
+
+    def _allreduce(self):
+        if self.ddp.world_size > 1:
+            handles = [
+                dist.all_reduce(b.buffer, op=dist.ReduceOp.SUM,
+                                group=self.ddp.group, async_op=True)
+                for b in self.ddp.buckets
+            ]
+            for h, b in zip(handles, self.ddp.buckets):
+                h.wait()
+                b.buffer.div_(self.ddp.world_size)
+
+    def initialize(self, batch: Dict[str, torch.Tensor]) -> None:
+        """Allocate static buffers, run eager warmups, capture both graphs.
+        Must be called with hooks disabled for world>1 (we all-reduce
+        manually between the graphs)."""
+        self.ddp.sync_enabled = False  # we all-reduce between the graphs
+        for k in ("x", "x_mask", "y", "y_mask"):
+            self.buf[k] = batch[k].clone()
+        # Warmup (MIOpen algo search, allocator, optimizer state init) on a
+        # side stream, the torch.cuda.graph idiom.
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                self._fwd_bwd()
+                self._allreduce()
+                self._clip_opt()
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+
+        self.g_fwdbwd = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.g_fwdbwd):
+            self.loss = self._fwd_bwd()
+        self.g_opt = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.g_opt):
+            self._clip_opt()
+
+    def step(self, batch: Optional[Dict[str, torch.Tensor]] = None) -> torch.Tensor:
+        """One training step; returns the (device) loss tensor."""
+        if batch is not None:
+            for k in self.buf:
+                self.buf[k].copy_(batch[k], non_blocking=True)
+        self.g_fwdbwd.replay()
+        self._allreduce()
+        self.g_opt.replay()
+        return self.loss
