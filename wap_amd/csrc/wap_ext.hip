@@ -160,7 +160,7 @@ __global__ void attn_score_bwd_kernel(
 // --------------------------------- masked softmax over L + context reduce
 // One workgroup (256 threads) per image.  alpha staged in LDS when L fits.
 
-template <typename T, bool LDS_ALPHA>
+template <typename T, bool LDS_ALPHA, bool WITH_CTX = true>
 __global__ void msc_fwd_kernel(
     const float* __restrict__ e, const T* __restrict__ a,
     const float* __restrict__ mask, float* __restrict__ alpha,
@@ -196,12 +196,33 @@ __global__ void msc_fwd_kernel(
     if (LDS_ALPHA) alpha[l] = v;
   }
   __syncthreads();
-  // context: lanes own consecutive d (coalesced over a's fastest dim)
-  for (int d = threadIdx.x; d < D; d += blockDim.x) {
-    float acc = 0.f;
-    for (int l = 0; l < L; ++l) acc += al[l] * ldf(a + (size_t)l * D + d);
-    ctx[d] = acc;
+  if (WITH_CTX) {
+    // context: lanes own consecutive d (coalesced over a's fastest dim)
+    for (int d = threadIdx.x; d < D; d += blockDim.x) {
+      float acc = 0.f;
+      for (int l = 0; l < L; ++l) acc += al[l] * ldf(a + (size_t)l * D + d);
+      ctx[d] = acc;
+    }
   }
+}
+
+// context reduction with a parallel grid (D-chunks, B); alpha staged in LDS.
+template <typename T>
+__global__ void msc_ctx_kernel(
+    const float* __restrict__ alpha, const T* __restrict__ a,
+    float* __restrict__ ctx, int L, int D) {
+  extern __shared__ float smem[];
+  const int b = blockIdx.y;
+  alpha += (size_t)b * L;
+  a += (size_t)b * L * D;
+  ctx += (size_t)b * D;
+  for (int l = threadIdx.x; l < L; l += blockDim.x) smem[l] = alpha[l];
+  __syncthreads();
+  const int d = blockIdx.x * blockDim.x + threadIdx.x;
+  if (d >= D) return;
+  float acc = 0.f;
+  for (int l = 0; l < L; ++l) acc += smem[l] * ldf(a + (size_t)l * D + d);
+  ctx[d] = acc;
 }
 
 // backward (two launches, grid (l-chunks, B) each):
@@ -511,14 +532,29 @@ std::vector<torch::Tensor> masked_softmax_context_fwd(
   auto ctx = torch::empty({B, D}, e.options());
   const bool lds = L <= 16384;
   const int smem = (16 + (lds ? L : 0)) * sizeof(float);
-#define LAUNCH_MSC(T, LDS) \
-  hipLaunchKernelGGL((msc_fwd_kernel<T, LDS>), dim3(B), dim3(256), smem, \
+#define LAUNCH_MSC(T, LDS, CTX) \
+  hipLaunchKernelGGL((msc_fwd_kernel<T, LDS, CTX>), dim3(B), dim3(256), smem, \
                      cur_stream(), cptr<float>(e), cptr<T>(a), \
                      cptr<float>(mask), ptr<float>(alpha), ptr<float>(ctx), L, D)
-  if (is_bf16(a)) {
-    if (lds) { LAUNCH_MSC(__hip_bfloat16, true); } else { LAUNCH_MSC(__hip_bfloat16, false); }
+  if (lds) {
+    // softmax-only pass, then the context reduction on a (D-chunks, B) grid
+    if (is_bf16(a)) { LAUNCH_MSC(__hip_bfloat16, true, false); }
+    else            { LAUNCH_MSC(float, true, false); }
+    HIP_OK();
+    const dim3 cgrid((D + 255) / 256, B);
+    const int csmem = L * sizeof(float);
+    if (is_bf16(a)) {
+      hipLaunchKernelGGL(msc_ctx_kernel<__hip_bfloat16>, cgrid, dim3(256),
+                         csmem, cur_stream(), cptr<float>(alpha),
+                         cptr<__hip_bfloat16>(a), ptr<float>(ctx), L, D);
+    } else {
+      hipLaunchKernelGGL(msc_ctx_kernel<float>, cgrid, dim3(256), csmem,
+                         cur_stream(), cptr<float>(alpha), cptr<float>(a),
+                         ptr<float>(ctx), L, D);
+    }
   } else {
-    if (lds) { LAUNCH_MSC(float, true); } else { LAUNCH_MSC(float, false); }
+    if (is_bf16(a)) { LAUNCH_MSC(__hip_bfloat16, false, true); }
+    else            { LAUNCH_MSC(float, false, true); }
   }
 #undef LAUNCH_MSC
   HIP_OK();

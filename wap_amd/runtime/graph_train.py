@@ -81,6 +81,10 @@ class GraphedTrainStep:
         Must be called with hooks disabled for world>1 (we all-reduce
         manually between the graphs)."""
         self.ddp.sync_enabled = False  # we all-reduce between the graphs
+        # torch optimizers require capturable=True (device-side step state)
+        # to be captured into a graph
+        for g in self.opt.param_groups:
+            g["capturable"] = True
         for k in ("x", "x_mask", "y", "y_mask"):
             self.buf[k] = batch[k].clone()
         # Warmup (MIOpen algo search, allocator, optimizer state init) on a
