@@ -47,6 +47,8 @@ def main() -> None:
     p.add_argument("--decode-steps", type=int, default=0,
                    help="decode measurement rounds (default 1 on GPU, 0 on CPU)")
     p.add_argument("--no-decode", action="store_true")
+    p.add_argument("--channels-last", type=str, default="on",
+                   choices=["on", "off"])
     p.add_argument("--graph", type=str, default="auto",
                    choices=["auto", "on", "off"],
                    help="hipGraph-captured train step")
@@ -69,6 +71,8 @@ def main() -> None:
         return torch.autocast("cuda", dtype=torch.bfloat16, enabled=use_bf16)
 
     model = WAPModel(n_tokens=112, encoder=args.encoder).to(device)
+    if on_gpu and args.channels_last == "on":
+        model.use_channels_last()
     ddp = BucketedDDP(model)
     opt = torch.optim.Adadelta(model.parameters(), lr=1.0, rho=0.95,
                                eps=1e-8, weight_decay=1e-4)
@@ -178,6 +182,7 @@ def main() -> None:
                 "parallelism": f"dp{world}",
                 "kernels": ops.get_backend(),
                 "graph": gstep is not None,
+                "channels_last": on_gpu and args.channels_last == "on",
             },
             "extra": extra,
         }))

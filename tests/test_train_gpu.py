@@ -167,3 +167,22 @@ def test_graphed_train_step_matches_eager():
     assert abs(graph_losses[-1] - eager_losses[4]) < 5e-3, \
         (graph_losses, eager_losses)
     assert graph_losses[-1] < eager_losses[0]
+
+
+def test_channels_last_matches_contiguous():
+    """NHWC encoder path must match NCHW numerics (fp32)."""
+    from wap_amd import ops
+    from wap_amd.models import WAPModel
+    ops.set_backend("auto")
+    torch.manual_seed(12)
+    dev = torch.device("cuda", 0)
+    m = WAPModel(n_tokens=30, encoder="densenet",
+                 encoder_kw={"growth": 8, "layers_per_block": 2}).to(dev).eval()
+    x = torch.rand(2, 1, 64, 96, device=dev)
+    xm = torch.ones(2, 64, 96, device=dev)
+    with torch.no_grad():
+        a1, _ = m.encode(x, xm)
+        m.use_channels_last()
+        a2, _ = m.encode(x, xm)
+    err = (a1 - a2).abs().max().item()
+    assert err < 1e-3, err

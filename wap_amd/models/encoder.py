@@ -111,6 +111,8 @@ class DenseNetEncoder(nn.Module):
 
     def forward(self, x: Tensor, x_mask: Tensor) -> Tuple[Tensor, Tensor]:
         h = conv_1ch(x, self.stem.weight, stride=2, padding=3)
+        if x.is_contiguous(memory_format=torch.channels_last):
+            h = h.to(memory_format=torch.channels_last)
         h = F.relu(self.stem_bn(h))
         h = F.max_pool2d(h, 2, ceil_mode=True)   # stride 4 after stem+pool
         a = self.blocks(h)

@@ -43,7 +43,18 @@ class WAPModel(nn.Module):
         )
         self.n_tokens = n_tokens
 
+    def use_channels_last(self) -> "WAPModel":
+        """Run the encoder in NHWC (channels_last): MIOpen's bf16 conv fast
+        paths are NHWC igemm kernels and otherwise transpose around every
+        conv (measured batched_transpose_* kernels); NHWC also makes the
+        attention's [B,L,D] annotation view copy-free."""
+        self._channels_last = True
+        self.watcher.to(memory_format=torch.channels_last)
+        return self
+
     def encode(self, x: Tensor, x_mask: Tensor) -> Tuple[Tensor, Tensor]:
+        if getattr(self, "_channels_last", False):
+            x = x.to(memory_format=torch.channels_last)
         return self.watcher(x, x_mask)
 
     def forward(
