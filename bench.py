@@ -47,7 +47,7 @@ def main() -> None:
     p.add_argument("--decode-steps", type=int, default=0,
                    help="decode measurement rounds (default 1 on GPU, 0 on CPU)")
     p.add_argument("--no-decode", action="store_true")
-    p.add_argument("--channels-last", type=str, default="on",
+    p.add_argument("--channels-last", type=str, default="off",
                    choices=["on", "off"])
     p.add_argument("--graph", type=str, default="auto",
                    choices=["auto", "on", "off"],

@@ -21,7 +21,7 @@ setup(
     ext_modules=[
         CUDAExtension(
             name="wap_amd._wap_hip",
-            sources=["wap_amd/csrc/wap_ext.hip"],
+            sources=["wap_amd/csrc/wap_ext.hip", "wap_amd/csrc/conv_mfma.hip"],
             extra_compile_args={
                 "cxx": ["-O3"],
                 "nvcc": ["-O3", "-std=c++17"],

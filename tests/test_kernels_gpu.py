@@ -184,3 +184,62 @@ def test_im2col_col2im_1ch(stride, pad, k, H, W, dtype):
     ref2 = torch.nn.functional.unfold(x2, (k, k), padding=pad, stride=stride)
     (dx_ref,) = torch.autograd.grad(ref2, x2, g.float())
     assert _rel_err(dx, dx_ref) < (1e-6 if dtype == torch.float32 else 1e-2)
+
+
+@pytest.mark.parametrize("C,K,R,HW", [(48, 96, 1, (16, 24)), (96, 24, 3, (16, 24)),
+                                      (300, 96, 1, (9, 13)), (96, 24, 3, (7, 11))])
+def test_bnrelu_conv_mfma_vs_conv2d(C, K, R, HW):
+    """MFMA implicit-GEMM fused conv vs relu(scale*x+shift) -> conv2d
+    (fp32 oracle, asymmetric weights: transpose-detecting per §5.4 r16)."""
+    from wap_amd.ops import ext
+    H, W = HW
+    dev = "cuda"
+    torch.manual_seed(7)
+    x = (torch.randn(3, C, H, W, device=dev) * 0.5).to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last)
+    w = (torch.randn(K, C, R, R, device=dev) * (1.0 / C ** 0.5)) \
+        .to(torch.bfloat16).contiguous(memory_format=torch.channels_last)
+    scale = torch.rand(C, device=dev) + 0.5
+    shift = torch.randn(C, device=dev) * 0.3
+    out = ext.require().bnrelu_conv_fwd(x, w, scale, shift, True)
+    xn = torch.relu(x.float() * scale.view(1, -1, 1, 1) + shift.view(1, -1, 1, 1))
+    ref = torch.nn.functional.conv2d(xn, w.float(), padding=R // 2)
+    assert out.shape == ref.shape
+    assert _rel_err(out, ref) < 2e-2, _rel_err(out, ref)
+
+
+def test_fused_bottleneck_grads_match_eager():
+    """Full fused BN-ReLU-conv autograd vs the eager composition."""
+    from wap_amd.ops.fused_conv import bnrelu_conv
+    dev = "cuda"
+    torch.manual_seed(8)
+    C, K = 48, 96
+    bn = torch.nn.BatchNorm2d(C).to(dev)
+    bn.weight.data.uniform_(0.5, 1.5)
+    bn.bias.data.normal_(0, 0.3)
+    conv = torch.nn.Conv2d(C, K, 1, bias=False).to(dev)
+    x0 = torch.randn(2, C, 12, 20, device=dev)
+
+    # eager fp32 reference
+    bn.train()
+    x1 = x0.clone().requires_grad_()
+    ref = conv(torch.relu(bn(x1)))
+    ref.square().sum().backward()
+    ref_dx, ref_dw = x1.grad.clone(), conv.weight.grad.clone()
+    ref_dg, ref_db = bn.weight.grad.clone(), bn.bias.grad.clone()
+    rm_ref = bn.running_mean.clone()
+
+    # fused path (bf16 channels_last)
+    for p in (conv.weight, bn.weight, bn.bias):
+        p.grad = None
+    bn.running_mean.zero_(); bn.running_var.fill_(1.0)
+    x2 = x0.clone().to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last).requires_grad_()
+    out = bnrelu_conv(x2, conv.weight, bn)
+    out.float().square().sum().backward()
+    assert _rel_err(out, ref) < 4e-2, _rel_err(out, ref)
+    assert _rel_err(x2.grad, ref_dx) < 6e-2
+    assert _rel_err(conv.weight.grad, ref_dw) < 6e-2
+    assert _rel_err(bn.weight.grad, ref_dg) < 6e-2
+    assert _rel_err(bn.bias.grad, ref_db) < 6e-2
+    assert _rel_err(bn.running_mean, rm_ref) < 1e-2

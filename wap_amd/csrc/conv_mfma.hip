@@ -1,0 +1,171 @@
+// MFMA implicit-GEMM convolution for the DenseNet bottleneck layers
+// (gfx950 / CDNA4), NHWC layout, bf16 operands, f32 accumulation, with the
+// pre-activation BN + ReLU fused into the input load stage.
+//
+// GEMM view (stride-1 "same" conv, NHWC):
+//   out[p, k] = sum_{(r,s), c} relu(x[p + D(r,s), c] * scale_c + shift_c)
+//               * w[k, r, s, c]
+// M = B*H*W spatial positions, N = K_out (<= 96 here: bottleneck convs are
+// C->96 (1x1) and 96->24 (3x3)), K-dim = R*S*C.
+//
+// Tiling: 256-thread workgroups (4 waves), BM=128 positions x BN=K_out,
+// BK=32 k-dim chunk = one v_mfma_f32_16x16x32_bf16 step; wave w owns rows
+// [w*32, w*32+32) => 2 x (BN/16) accumulator fragments.  A (input) and B
+// (weight) tiles are LDS-staged; BN/ReLU applied on the way in.
+// Correctness-first: single-buffered, two barriers per K-chunk.  See
+// profiles/ for measured utilization.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#define WAVE 64
+#define DEVC __device__ __forceinline__
+
+typedef __hip_bfloat16 bf16_t;
+typedef __attribute__((ext_vector_type(8))) short bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+DEVC float bf2f(bf16_t v) { return __bfloat162float(v); }
+DEVC bf16_t f2bf(float v) { return __float2bfloat16(v); }
+
+// ---------------------------------------------------------------------
+// Fused BN-ReLU-conv forward.
+//   x     [B, H, W, C]   bf16 (NHWC physical)
+//   w     [K, R, S, C]   bf16 (channels_last physical)
+//   scale [C], shift [C] f32  (BN folded: scale=gamma*rsqrt(var+eps),
+//                              shift=beta-mean*scale; identity: 1/0)
+//   out   [B, H, W, K]   bf16
+// Grid: (ceil(BHW/BM), 1); 256 threads.  KOUT <= 96, C % 8 == 0.
+// ---------------------------------------------------------------------
+
+#define BM 128
+#define BKC 32   // kdim chunk (channels per MFMA step)
+
+extern "C" __global__ __launch_bounds__(256, 2)
+void bnrelu_conv_mfma_fwd(
+    const bf16_t* __restrict__ x, const bf16_t* __restrict__ w,
+    const float* __restrict__ scale, const float* __restrict__ shift,
+    bf16_t* __restrict__ out,
+    int Bn, int H, int W, int C, int KOUT, int R, int S, int relu_on) {
+  // LDS: A tile [BM][BKC] bf16 (8 KB) + B tile [96][BKC] bf16 (6 KB)
+  __shared__ bf16_t lA[BM][BKC];
+  __shared__ bf16_t lB[96][BKC];
+  __shared__ float lscale[BKC], lshift[BKC];
+
+  const int HW = H * W;
+  const long M = (long)Bn * HW;
+  const long m0 = (long)blockIdx.x * BM;
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wv = tid >> 6;               // wave id 0..3
+  const int PR = R / 2, PS = S / 2;      // "same" padding
+
+  const int NFRAG = (KOUT + 15) / 16;    // <= 6
+  f32x4 acc[2][6];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 6; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  for (int r = 0; r < R; ++r) {
+    for (int s = 0; s < S; ++s) {
+      for (int c0 = 0; c0 < C; c0 += BKC) {
+        // ---- stage scale/shift for this channel chunk
+        if (tid < BKC) {
+          const int c = c0 + tid;
+          lscale[tid] = c < C ? scale[c] : 0.f;
+          lshift[tid] = c < C ? shift[c] : 0.f;
+        }
+        __syncthreads();
+        // ---- stage A: BM positions x BKC channels; each thread stages
+        // 2 of the 512 16-byte (8 x bf16) pieces
+#pragma unroll
+        for (int i = 0; i < 2; ++i) {
+          const int flat = tid * 2 + i;        // 0..511
+          const int arow = flat >> 2;          // 0..127
+          const int apc = flat & 3;            // 16B piece in row
+          const long p = m0 + arow;
+          float vals[8];
+          if (p < M) {
+            const int b = (int)(p / HW);
+            const int rem = (int)(p % HW);
+            const int y = rem / W + r - PR;
+            const int xw = rem % W + s - PS;
+            if (y >= 0 && y < H && xw >= 0 && xw < W) {
+              const long base = (((long)b * H + y) * W + xw) * C + c0 + apc * 8;
+#pragma unroll
+              for (int e = 0; e < 8; ++e) {
+                const int c = c0 + apc * 8 + e;
+                vals[e] = (c < C) ? bf2f(x[base + e]) : 0.f;
+              }
+            } else {
+#pragma unroll
+              for (int e = 0; e < 8; ++e) vals[e] = 0.f;
+            }
+          } else {
+#pragma unroll
+            for (int e = 0; e < 8; ++e) vals[e] = 0.f;
+          }
+          // BN + ReLU on the way into LDS
+#pragma unroll
+          for (int e = 0; e < 8; ++e) {
+            const int cc = apc * 8 + e;
+            float v = vals[e] * lscale[cc] + lshift[cc];
+            if (relu_on) v = fmaxf(v, 0.f);
+            lA[arow][cc] = f2bf(v);
+          }
+        }
+        // ---- stage B: KOUT rows x BKC channels of tap (r,s)
+        // thread t stages pieces of w[k, r, s, c0..c0+31]
+        for (int flat = tid; flat < NFRAG * 16 * 4; flat += blockDim.x) {
+          const int k = flat >> 2;
+          const int bpc = flat & 3;
+          if (k < 96) {
+#pragma unroll
+            for (int e = 0; e < 8; ++e) {
+              const int c = c0 + bpc * 8 + e;
+              float v = 0.f;
+              if (k < KOUT && c < C)
+                v = bf2f(w[(((long)k * R + r) * S + s) * C + c]);
+              lB[k][bpc * 8 + e] = f2bf(v);
+            }
+          }
+        }
+        __syncthreads();
+        // ---- MFMA: wave wv covers rows [wv*32, wv*32+32)
+#pragma unroll
+        for (int mf = 0; mf < 2; ++mf) {
+          const int arow = wv * 32 + mf * 16 + (lane & 15);
+          const int kk = (lane >> 4) * 8;
+          bf16x8 afrag = *(const bf16x8*)&lA[arow][kk];
+#pragma unroll
+          for (int nf = 0; nf < 6; ++nf) {
+            if (nf >= NFRAG) break;
+            const int brow = nf * 16 + (lane & 15);
+            bf16x8 bfrag = *(const bf16x8*)&lB[brow][kk];
+            acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afrag, bfrag, acc[mf][nf], 0, 0, 0);
+          }
+        }
+        __syncthreads();
+      }
+    }
+  }
+
+  // ---- epilogue: C/D layout col=lane&15 (out channel), row=(lane>>4)*4+reg
+#pragma unroll
+  for (int mf = 0; mf < 2; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 6; ++nf) {
+      if (nf >= NFRAG) break;
+      const int k = nf * 16 + (lane & 15);
+      if (k >= KOUT) continue;
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int mrow = wv * 32 + mf * 16 + (lane >> 4) * 4 + reg;
+        const long p = m0 + mrow;
+        if (p < M) out[p * KOUT + k] = f2bf(acc[mf][nf][reg]);
+      }
+    }
+  }
+}

@@ -723,7 +723,41 @@ torch::Tensor fused_gru_cell(
   return h_new;
 }
 
+// MFMA implicit-GEMM conv (conv_mfma.hip)
+extern "C" __global__ void bnrelu_conv_mfma_fwd(
+    const __hip_bfloat16*, const __hip_bfloat16*, const float*, const float*,
+    __hip_bfloat16*, int, int, int, int, int, int, int, int);
+
+torch::Tensor bnrelu_conv_fwd(
+    torch::Tensor x, torch::Tensor w, torch::Tensor scale,
+    torch::Tensor shift, bool relu_on) {
+  TORCH_CHECK(x.is_cuda() && w.is_cuda());
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16 && w.scalar_type() == at::kBFloat16,
+              "bnrelu_conv: bf16 only");
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "x must be channels_last (NHWC)");
+  TORCH_CHECK(w.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "w must be channels_last");
+  TORCH_CHECK(scale.scalar_type() == at::kFloat && shift.scalar_type() == at::kFloat);
+  const int B = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  const int K = w.size(0), R = w.size(2), S = w.size(3);
+  TORCH_CHECK(w.size(1) == C && K <= 96 && (R % 2) && (S % 2),
+              "bnrelu_conv: K<=96, odd kernel, same channels");
+  auto out = at::empty({B, K, H, W}, x.options(),
+                       at::MemoryFormat::ChannelsLast);
+  const long M = (long)B * H * W;
+  const dim3 grid((M + 127) / 128);
+  hipLaunchKernelGGL(bnrelu_conv_mfma_fwd, grid, dim3(256), 0, cur_stream(),
+                     cptr<__hip_bfloat16>(x), cptr<__hip_bfloat16>(w),
+                     cptr<float>(scale), cptr<float>(shift),
+                     ptr<__hip_bfloat16>(out), B, H, W, C, K, R, S,
+                     relu_on ? 1 : 0);
+  HIP_OK();
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("bnrelu_conv_fwd", &bnrelu_conv_fwd);
   m.def("attention_score_fwd", &attention_score_fwd);
   m.def("attention_score_bwd", &attention_score_bwd);
   m.def("masked_softmax_context_fwd", &masked_softmax_context_fwd);

@@ -66,8 +66,15 @@ class _Bottleneck(nn.Module):
         self.dropout = dropout
 
     def forward(self, x: Tensor) -> Tensor:
-        out = self.conv1(F.relu(self.bn1(x)))
-        out = self.conv2(F.relu(self.bn2(out)))
+        from ..ops import fused_conv
+        if fused_conv.can_fuse(x, self.conv1.weight):
+            # hand-written MFMA implicit-GEMM conv with BN+ReLU fused into
+            # the load stage (csrc/conv_mfma.hip)
+            out = fused_conv.bnrelu_conv(x, self.conv1.weight, self.bn1)
+            out = fused_conv.bnrelu_conv(out, self.conv2.weight, self.bn2)
+        else:
+            out = self.conv1(F.relu(self.bn1(x)))
+            out = self.conv2(F.relu(self.bn2(out)))
         if self.dropout > 0 and self.training:
             out = F.dropout(out, self.dropout)
         return torch.cat([x, out], dim=1)
