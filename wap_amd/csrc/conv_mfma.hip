@@ -85,34 +85,43 @@ void bnrelu_conv_mfma_fwd(
           const int arow = flat >> 2;          // 0..127
           const int apc = flat & 3;            // 16B piece in row
           const long p = m0 + arow;
-          float vals[8];
+          // Conv zero-padding applies AFTER the BN+ReLU (the reference
+          // composition pads the normalized tensor): out-of-bounds taps
+          // contribute exactly 0, never relu(shift).
+          bool inb = false;
+          long base = 0;
           if (p < M) {
             const int b = (int)(p / HW);
             const int rem = (int)(p % HW);
             const int y = rem / W + r - PR;
             const int xw = rem % W + s - PS;
             if (y >= 0 && y < H && xw >= 0 && xw < W) {
-              const long base = (((long)b * H + y) * W + xw) * C + c0 + apc * 8;
+              inb = true;
+              base = (((long)b * H + y) * W + xw) * C + c0 + apc * 8;
+            }
+          }
+          if (inb && c0 + apc * 8 + 8 <= C) {
+            // fast path: aligned 16-byte load + BN/ReLU
+            const bf16x8 v8 = *(const bf16x8*)(x + base);
 #pragma unroll
-              for (int e = 0; e < 8; ++e) {
-                const int c = c0 + apc * 8 + e;
-                vals[e] = (c < C) ? bf2f(x[base + e]) : 0.f;
-              }
-            } else {
-#pragma unroll
-              for (int e = 0; e < 8; ++e) vals[e] = 0.f;
+            for (int e = 0; e < 8; ++e) {
+              const int cc = apc * 8 + e;
+              float v = bf2f(((const bf16_t*)&v8)[e]) * lscale[cc] + lshift[cc];
+              if (relu_on) v = fmaxf(v, 0.f);
+              lA[arow][cc] = f2bf(v);
             }
           } else {
 #pragma unroll
-            for (int e = 0; e < 8; ++e) vals[e] = 0.f;
-          }
-          // BN + ReLU on the way into LDS
-#pragma unroll
-          for (int e = 0; e < 8; ++e) {
-            const int cc = apc * 8 + e;
-            float v = vals[e] * lscale[cc] + lshift[cc];
-            if (relu_on) v = fmaxf(v, 0.f);
-            lA[arow][cc] = f2bf(v);
+            for (int e = 0; e < 8; ++e) {
+              const int cc = apc * 8 + e;
+              const int c = c0 + cc;
+              float v = 0.f;
+              if (inb && c < C) {
+                v = bf2f(x[base + e]) * lscale[cc] + lshift[cc];
+                if (relu_on) v = fmaxf(v, 0.f);
+              }
+              lA[arow][cc] = f2bf(v);
+            }
           }
         }
         // ---- stage B: KOUT rows x BKC channels of tap (r,s)

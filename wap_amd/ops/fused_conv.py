@@ -82,6 +82,11 @@ class _BNReluConv(torch.autograd.Function):
 
 
 def can_fuse(x: Tensor, weight: Tensor) -> bool:
+    import os
+    if os.environ.get("WAP_FUSED_CONV", "0") != "1":
+        # Opt-in while the MFMA kernel is tuned up to MIOpen's igemm
+        # throughput (single-buffered v1; see PERF.md).
+        return False
     return (
         use_hip(x)
         and x.dtype == torch.bfloat16
