@@ -121,3 +121,26 @@ def test_on_disk_roundtrip(tmp_path, vocab):
     assert len(samples) == 12
     assert all(len(s.label) > 0 for s in samples)
     assert all(s.image.max() > 0 for s in samples)  # not zero-filled (DVFS rule)
+
+
+def test_pad_to_keeps_loss_exact():
+    """pad_to rounds shapes up + pads the batch with all-masked rows; with a
+    BN-free encoder (VGG) the masked loss must be bit-comparable."""
+    import torch
+    from wap_amd.models import WAPModel
+    torch.manual_seed(31)
+    m = WAPModel(n_tokens=15, encoder="vgg", embed_dim=16, hidden_dim=16,
+                 attention_dim=16, coverage_ch=8, coverage_kernel=3).eval()
+    samples = _mk_samples(3, seed=9, min_h=20, max_h=40, min_w=30, max_w=50)
+    for s in samples:  # clamp ids into the toy model's 15-token vocab
+        s.label = [1 + (i % 13) for i in s.label]
+    from wap_amd.data import Bucket, prepare_batch
+    plain = prepare_batch(Bucket(samples))
+    padded = prepare_batch(Bucket(samples), pad_to=(32, 32, 8, 6))
+    assert padded["x"].shape[0] == 6
+    assert padded["x"].shape[2] % 32 == 0 and padded["x"].shape[3] % 32 == 0
+    assert padded["y"].shape[1] % 8 == 0
+    with torch.no_grad():
+        l1 = m(plain["x"], plain["x_mask"], plain["y"], plain["y_mask"])["loss"]
+        l2 = m(padded["x"], padded["x_mask"], padded["y"], padded["y_mask"])["loss"]
+    assert torch.allclose(l1, l2, atol=1e-5), (l1.item(), l2.item())

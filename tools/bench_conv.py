@@ -2,7 +2,11 @@
 """Microbench: MFMA fused BN-ReLU-conv vs MIOpen (BN+ReLU+conv2d) on the
 DenseNet bottleneck shapes.  Run on the GPU box."""
 
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 

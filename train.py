@@ -135,11 +135,22 @@ def main() -> None:
         log.log("resume", path=cfg.resume, epoch=start_epoch, best_wer=best_wer)
 
     ddp = BucketedDDP(model, bucket_bytes=cfg.bucket_mb << 20)
-    loader = BucketLoader(
-        train_buckets, rank=rank, world_size=world, seed=cfg.seed, device=device
-    )
-
     use_bf16 = device.type == "cuda" and cfg.dtype == "bf16"
+    use_graph = device.type == "cuda" and cfg.graph != "off"
+    trainer = None
+    pad_to = None
+    if use_graph:
+        from wap_amd.runtime.graph_train import MultiShapeGraphedTrainer
+        trainer = MultiShapeGraphedTrainer(
+            model, ddp, optimizer, clip=cfg.grad_clip, autocast_bf16=use_bf16
+        )
+        # collapse bucket shapes onto a grid so captured graphs are reused
+        pad_to = (cfg.graph_pad_hw, cfg.graph_pad_hw, cfg.graph_pad_t,
+                  cfg.batch_size)
+    loader = BucketLoader(
+        train_buckets, rank=rank, world_size=world, seed=cfg.seed,
+        device=device, pad_to=pad_to,
+    )
 
     def autocast_ctx():
         return torch.autocast("cuda", dtype=torch.bfloat16, enabled=use_bf16)
@@ -152,17 +163,23 @@ def main() -> None:
         seen_imgs = 0
         running = 0.0
         for i, batch in enumerate(loader):
-            ddp.zero_grad()
-            with autocast_ctx():
-                out = model(batch["x"], batch["x_mask"], batch["y"], batch["y_mask"])
-            out["loss"].backward()
-            ddp.finish_gradient_sync()
-            if cfg.grad_clip > 0:
-                torch.nn.utils.clip_grad_norm_(model.parameters(), cfg.grad_clip)
-            optimizer.step()
+            if trainer is not None:
+                loss_t = trainer.step(batch)
+            else:
+                ddp.zero_grad()
+                with autocast_ctx():
+                    out = model(batch["x"], batch["x_mask"],
+                                batch["y"], batch["y_mask"])
+                out["loss"].backward()
+                ddp.finish_gradient_sync()
+                if cfg.grad_clip > 0:
+                    torch.nn.utils.clip_grad_norm_(model.parameters(),
+                                                   cfg.grad_clip)
+                optimizer.step()
+                loss_t = out["loss"]
             gstep += 1
             seen_imgs += batch["x"].shape[0]
-            running += out["loss"].item()
+            running += loss_t.item()
             if (i + 1) % cfg.log_every == 0:
                 dt = time.perf_counter() - t0
                 log.log("train", epoch=epoch, step=gstep,

@@ -48,6 +48,9 @@ class WAPConfig:
     device: str = "auto"               # auto | cpu | cuda
     dtype: str = "bf16"                # bf16 | fp32 (compute dtype on GPU)
     kernels: str = "auto"              # auto | eager | hip
+    graph: str = "auto"                # auto | off: hipGraph-captured steps
+    graph_pad_hw: int = 64             # bucket shape padding grid for graphs
+    graph_pad_t: int = 16
     log_every: int = 20
     valid_every_epochs: int = 1
     ckpt_dir: str = "checkpoints"

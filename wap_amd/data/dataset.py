@@ -28,7 +28,7 @@ from __future__ import annotations
 import pickle
 import random
 from dataclasses import dataclass, field
-from typing import Dict, Iterator, List, Sequence, Tuple
+from typing import Dict, Iterator, List, Optional, Sequence, Tuple
 
 import numpy as np
 import torch
@@ -138,10 +138,15 @@ def shard_buckets(buckets: List[Bucket], rank: int, world_size: int) -> List[Buc
     return mine[:n_steps]
 
 
+def _round_up(v: int, m: int) -> int:
+    return ((v + m - 1) // m) * m
+
+
 def prepare_batch(
     bucket: Bucket,
     device: torch.device | str = "cpu",
     dtype: torch.dtype = torch.float32,
+    pad_to: Optional[Tuple[int, int, int, int]] = None,
 ) -> Dict[str, torch.Tensor]:
     """Pad one bucket to common H×W / T and build masks.
 
@@ -149,11 +154,22 @@ def prepare_batch(
     ``y[B,T]`` with <eos> appended, ``y_mask[B,T]`` (1=real token incl. the
     eos position) — the same information as the reference's
     ``prepare_data`` but batch-major NCHW.
+
+    ``pad_to=(hm, wm, tm, bm)`` rounds H/W/T up to multiples and pads the
+    batch to ``bm`` rows (all-masked images), collapsing bucket shapes onto
+    a small grid so hipGraph-captured train steps can be reused across
+    buckets; masks keep the loss exact (see tests/test_data.py).
     """
     b = len(bucket.samples)
     hmax = max(s.image.shape[0] for s in bucket.samples)
     wmax = max(s.image.shape[1] for s in bucket.samples)
     tmax = max(len(s.label) for s in bucket.samples) + 1  # +1 for eos
+    if pad_to is not None:
+        hm, wm, tm, bm = pad_to
+        hmax = _round_up(hmax, hm)
+        wmax = _round_up(wmax, wm)
+        tmax = _round_up(tmax, tm)
+        b = max(b, bm)
 
     x = np.zeros((b, 1, hmax, wmax), dtype=np.float32)
     x_mask = np.zeros((b, hmax, wmax), dtype=np.float32)
@@ -192,12 +208,14 @@ class BucketLoader:
         seed: int = 0,
         device: torch.device | str = "cpu",
         dtype: torch.dtype = torch.float32,
+        pad_to: Optional[Tuple[int, int, int, int]] = None,
     ):
         self.all_buckets = buckets
         self.local = shard_buckets(buckets, rank, world_size)
         self.seed = seed
         self.device = device
         self.dtype = dtype
+        self.pad_to = pad_to
         self.epoch = 0
 
     def set_epoch(self, epoch: int) -> None:
@@ -210,4 +228,5 @@ class BucketLoader:
         order = list(range(len(self.local)))
         random.Random(self.seed * 1_000_003 + self.epoch).shuffle(order)
         for i in order:
-            yield prepare_batch(self.local[i], self.device, self.dtype)
+            yield prepare_batch(self.local[i], self.device, self.dtype,
+                                pad_to=self.pad_to)

@@ -102,6 +102,11 @@ class GraphedTrainStep:
         self.g_fwdbwd = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.g_fwdbwd):
             self.loss = self._fwd_bwd()
+        # Graph capture EXECUTES the captured ops: the fwd+bwd above left
+        # local (unreduced) grads in the buckets.  All-reduce them before
+        # capturing the optimizer graph — its capture-time execution applies
+        # a step, which must be rank-consistent.
+        self._allreduce()
         self.g_opt = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.g_opt):
             self._clip_opt()
@@ -115,3 +120,128 @@ class GraphedTrainStep:
         self._allreduce()
         self.g_opt.replay()
         return self.loss
+
+
+class _ShapeGraph:
+    """A captured fwd+bwd for one batch-shape: static input buffers + graph."""
+
+    def __init__(self, trainer: "MultiShapeGraphedTrainer",
+                 batch: Dict[str, torch.Tensor]):
+        self.t = trainer
+        self.buf = {k: batch[k].clone() for k in ("x", "x_mask", "y", "y_mask")}
+        # Warmup fwd+bwd only — NO collectives, NO optimizer: lazy capture
+        # must not disturb the per-step all-reduce alignment across ranks.
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            self._fwd_bwd()
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            self.loss = self._fwd_bwd()
+        # capture executed fwd+bwd on this batch: grads are now valid.
+
+    def _fwd_bwd(self):
+        for b in self.t.ddp.buckets:
+            b.buffer.zero_()
+        with torch.autocast("cuda", dtype=torch.bfloat16,
+                            enabled=self.t.autocast_bf16):
+            out = self.t.model(self.buf["x"], self.buf["x_mask"],
+                               self.buf["y"], self.buf["y_mask"])
+        out["loss"].backward()
+        return out["loss"]
+
+
+class MultiShapeGraphedTrainer:
+    """Graphed training over VARIABLE bucket shapes (train.py).
+
+    - fwd+bwd graphs are captured lazily per shape, with no collectives in
+      the capture path, so ranks may capture at different times without
+      breaking all-reduce alignment (each step still performs exactly one
+      bucketed all-reduce sweep + one optimizer step on every rank);
+    - the clip+optimizer graph is captured once, at the first step (all
+      ranks reach it together);
+    - shapes beyond ``max_graphs`` run an eager fwd+bwd fallback with the
+      same collective pattern.
+
+    Use together with ``prepare_batch(pad_to=...)`` so bucket shapes
+    collapse onto a small grid.
+    """
+
+    def __init__(self, model, ddp, optimizer, clip: float = 100.0,
+                 autocast_bf16: bool = True, max_graphs: int = 24):
+        self.model = model
+        self.ddp = ddp
+        self.opt = optimizer
+        self.clip = clip
+        self.autocast_bf16 = autocast_bf16
+        self.max_graphs = max_graphs
+        self.cache: Dict[tuple, _ShapeGraph] = {}
+        self.g_opt: Optional[torch.cuda.CUDAGraph] = None
+        self.n_eager_fallback = 0
+        ddp.sync_enabled = False
+        for g in self.opt.param_groups:
+            g["capturable"] = True
+
+    def _allreduce(self):
+        if self.ddp.world_size > 1:
+            handles = [
+                dist.all_reduce(b.buffer, op=dist.ReduceOp.SUM,
+                                group=self.ddp.group, async_op=True)
+                for b in self.ddp.buckets
+            ]
+            for h, b in zip(handles, self.ddp.buckets):
+                h.wait()
+                b.buffer.div_(self.ddp.world_size)
+
+    def _opt_step(self):
+        if self.g_opt is not None:
+            self.g_opt.replay()
+            return
+        # First step on every rank: capture the clip+opt graph (capture
+        # executes it once = this step's update).
+        torch.cuda.synchronize()
+        self.g_opt = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.g_opt):
+            if self.clip > 0:
+                torch.nn.utils.clip_grad_norm_(self.model.parameters(), self.clip)
+            self.opt.step()
+
+    def _eager_fwd_bwd(self, batch):
+        for b in self.ddp.buckets:
+            b.buffer.zero_()
+        with torch.autocast("cuda", dtype=torch.bfloat16,
+                            enabled=self.autocast_bf16):
+            out = self.model(batch["x"], batch["x_mask"],
+                             batch["y"], batch["y_mask"])
+        out["loss"].backward()
+        return out["loss"]
+
+    def step(self, batch: Dict[str, torch.Tensor]) -> torch.Tensor:
+        key = tuple(batch["x"].shape) + tuple(batch["y"].shape)
+        sg = self.cache.get(key)
+        if sg is not None:
+            for k in sg.buf:
+                sg.buf[k].copy_(batch[k], non_blocking=True)
+            sg.graph.replay()
+            loss = sg.loss
+        elif len(self.cache) < self.max_graphs:
+            # Optimizer state must exist before its graph can be captured;
+            # the Adadelta state dict is lazily created at the first eager
+            # step, so the opt graph is only captured from step 2 on.
+            sg = _ShapeGraph(self, batch)  # capture ran fwd+bwd on batch
+            self.cache[key] = sg
+            loss = sg.loss
+        else:
+            self.n_eager_fallback += 1
+            loss = self._eager_fwd_bwd(batch)
+        self._allreduce()
+        if self.g_opt is None and not self.opt.state:
+            # very first step: eager optimizer step initializes state
+            if self.clip > 0:
+                torch.nn.utils.clip_grad_norm_(self.model.parameters(), self.clip)
+            self.opt.step()
+        else:
+            self._opt_step()
+        return loss
