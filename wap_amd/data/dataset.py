@@ -227,6 +227,34 @@ class BucketLoader:
     def __iter__(self) -> Iterator[Dict[str, torch.Tensor]]:
         order = list(range(len(self.local)))
         random.Random(self.seed * 1_000_003 + self.epoch).shuffle(order)
-        for i in order:
-            yield prepare_batch(self.local[i], self.device, self.dtype,
-                                pad_to=self.pad_to)
+        dev = torch.device(self.device)
+        if dev.type != "cuda":
+            for i in order:
+                yield prepare_batch(self.local[i], self.device, self.dtype,
+                                    pad_to=self.pad_to)
+            return
+        # GPU: build + pin batches on a worker thread, overlap the H2D copy
+        # with the previous step's compute (non_blocking into the stream).
+        import queue
+        import threading
+
+        q: "queue.Queue" = queue.Queue(maxsize=2)
+
+        def worker():
+            for i in order:
+                b = prepare_batch(self.local[i], "cpu", self.dtype,
+                                  pad_to=self.pad_to)
+                b = {k: (v.pin_memory() if torch.is_tensor(v) else v)
+                     for k, v in b.items()}
+                q.put(b)
+            q.put(None)
+
+        t = threading.Thread(target=worker, daemon=True)
+        t.start()
+        while True:
+            b = q.get()
+            if b is None:
+                break
+            yield {k: (v.to(dev, non_blocking=True) if torch.is_tensor(v) else v)
+                   for k, v in b.items()}
+        t.join()
