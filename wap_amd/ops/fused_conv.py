@@ -83,9 +83,13 @@ class _BNReluConv(torch.autograd.Function):
 
 def can_fuse(x: Tensor, weight: Tensor) -> bool:
     import os
-    if os.environ.get("WAP_FUSED_CONV", "0") != "1":
-        # Opt-in while the MFMA kernel is tuned up to MIOpen's igemm
-        # throughput (single-buffered v1; see PERF.md).
+    mode = os.environ.get("WAP_FUSED_CONV", "1x1")
+    if mode == "0":
+        return False
+    if mode == "1x1" and weight.shape[2] != 1:
+        # measured (tools/bench_conv.py): the MFMA kernel beats MIOpen
+        # 2-3x on the 1x1 bottleneck convs, but the single-buffered 3x3
+        # path still trails igemm -- 3x3 stays on MIOpen until tuned.
         return False
     return (
         use_hip(x)

@@ -137,7 +137,7 @@ class BeamState:
         torch.cuda.synchronize()
         g = torch.cuda.CUDAGraph()
         ctx = torch.autocast("cuda", dtype=torch.bfloat16, enabled=autocast_bf16)
-        with ctx, torch.cuda.graph(g):
+        with ctx, torch.cuda.graph(g, capture_error_mode="thread_local"):
             _beam_step_(self.model, self.S, self.b, self.k, fused)
         self.graph = g
 

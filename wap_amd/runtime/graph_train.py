@@ -100,7 +100,7 @@ class GraphedTrainStep:
         torch.cuda.synchronize()
 
         self.g_fwdbwd = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.g_fwdbwd):
+        with torch.cuda.graph(self.g_fwdbwd, capture_error_mode="thread_local"):
             self.loss = self._fwd_bwd()
         # Graph capture EXECUTES the captured ops: the fwd+bwd above left
         # local (unreduced) grads in the buckets.  All-reduce them before
@@ -108,7 +108,7 @@ class GraphedTrainStep:
         # a step, which must be rank-consistent.
         self._allreduce()
         self.g_opt = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.g_opt):
+        with torch.cuda.graph(self.g_opt, capture_error_mode="thread_local"):
             self._clip_opt()
 
     def step(self, batch: Optional[Dict[str, torch.Tensor]] = None) -> torch.Tensor:
@@ -138,7 +138,7 @@ class _ShapeGraph:
         torch.cuda.current_stream().wait_stream(s)
         torch.cuda.synchronize()
         self.graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.graph):
+        with torch.cuda.graph(self.graph, capture_error_mode="thread_local"):
             self.loss = self._fwd_bwd()
         # capture executed fwd+bwd on this batch: grads are now valid.
 
@@ -178,6 +178,7 @@ class MultiShapeGraphedTrainer:
         self.autocast_bf16 = autocast_bf16
         self.max_graphs = max_graphs
         self.cache: Dict[tuple, _ShapeGraph] = {}
+        self._failed: set = set()
         self.g_opt: Optional[torch.cuda.CUDAGraph] = None
         self.n_eager_fallback = 0
         ddp.sync_enabled = False
@@ -203,7 +204,7 @@ class MultiShapeGraphedTrainer:
         # executes it once = this step's update).
         torch.cuda.synchronize()
         self.g_opt = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.g_opt):
+        with torch.cuda.graph(self.g_opt, capture_error_mode="thread_local"):
             if self.clip > 0:
                 torch.nn.utils.clip_grad_norm_(self.model.parameters(), self.clip)
             self.opt.step()
@@ -226,13 +227,20 @@ class MultiShapeGraphedTrainer:
                 sg.buf[k].copy_(batch[k], non_blocking=True)
             sg.graph.replay()
             loss = sg.loss
-        elif len(self.cache) < self.max_graphs:
+        elif len(self.cache) < self.max_graphs and key not in self._failed:
             # Optimizer state must exist before its graph can be captured;
             # the Adadelta state dict is lazily created at the first eager
             # step, so the opt graph is only captured from step 2 on.
-            sg = _ShapeGraph(self, batch)  # capture ran fwd+bwd on batch
-            self.cache[key] = sg
-            loss = sg.loss
+            try:
+                sg = _ShapeGraph(self, batch)  # capture ran fwd+bwd on batch
+                self.cache[key] = sg
+                loss = sg.loss
+            except Exception as e:
+                import sys
+                print(f"[graph_train] capture failed for {key}: {e!r}; "
+                      "eager fallback", file=sys.stderr)
+                self._failed.add(key)
+                loss = self._eager_fwd_bwd(batch)
         else:
             self.n_eager_fallback += 1
             loss = self._eager_fwd_bwd(batch)
