@@ -727,6 +727,9 @@ torch::Tensor fused_gru_cell(
 extern "C" __global__ void bnrelu_conv_mfma_fwd(
     const __hip_bfloat16*, const __hip_bfloat16*, const float*, const float*,
     __hip_bfloat16*, int, int, int, int, int, int, int, int);
+extern "C" __global__ void bnrelu_conv_mfma_fwd_bk96(
+    const __hip_bfloat16*, const __hip_bfloat16*, const float*, const float*,
+    __hip_bfloat16*, int, int, int, int, int, int, int, int);
 
 torch::Tensor bnrelu_conv_fwd(
     torch::Tensor x, torch::Tensor w, torch::Tensor scale,
@@ -747,7 +750,8 @@ torch::Tensor bnrelu_conv_fwd(
                        at::MemoryFormat::ChannelsLast);
   const long M = (long)B * H * W;
   const dim3 grid((M + 127) / 128);
-  hipLaunchKernelGGL(bnrelu_conv_mfma_fwd, grid, dim3(256), 0, cur_stream(),
+  auto kern = (C % 96 == 0) ? bnrelu_conv_mfma_fwd_bk96 : bnrelu_conv_mfma_fwd;
+  hipLaunchKernelGGL(kern, grid, dim3(256), 0, cur_stream(),
                      cptr<__hip_bfloat16>(x), cptr<__hip_bfloat16>(w),
                      cptr<float>(scale), cptr<float>(shift),
                      ptr<__hip_bfloat16>(out), B, H, W, C, K, R, S,
