@@ -24,8 +24,6 @@ import time
 
 import torch
 
-torch.backends.cudnn.benchmark = True  # MIOpen exhaustive find
-
 from wap_amd import ops
 from wap_amd.data.synth import fixed_shape_batch
 from wap_amd.models import WAPModel

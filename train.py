@@ -24,8 +24,6 @@ import time
 
 import torch
 
-torch.backends.cudnn.benchmark = True  # MIOpen exhaustive find
-
 from wap_amd import ops
 from wap_amd.config import WAPConfig, add_config_args, config_from_args
 from wap_amd.data import (
