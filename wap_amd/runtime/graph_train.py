@@ -102,10 +102,9 @@ class GraphedTrainStep:
         self.g_fwdbwd = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.g_fwdbwd, capture_error_mode="thread_local"):
             self.loss = self._fwd_bwd()
-        # Graph capture EXECUTES the captured ops: the fwd+bwd above left
-        # local (unreduced) grads in the buckets.  All-reduce them before
-        # capturing the optimizer graph — its capture-time execution applies
-        # a step, which must be rank-consistent.
+        # (Stream capture records without executing; grads in the buckets
+        # are still the last warmup's.  The all-reduce keeps them finite and
+        # rank-consistent for the opt-graph recording's tensor metadata.)
         self._allreduce()
         self.g_opt = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.g_opt, capture_error_mode="thread_local"):
@@ -140,7 +139,8 @@ class _ShapeGraph:
         self.graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.graph, capture_error_mode="thread_local"):
             self.loss = self._fwd_bwd()
-        # capture executed fwd+bwd on this batch: grads are now valid.
+        # NOTE: stream capture RECORDS the ops without executing them — the
+        # caller must replay() to actually compute this batch.
 
     def _fwd_bwd(self):
         for b in self.t.ddp.buckets:
@@ -232,8 +232,9 @@ class MultiShapeGraphedTrainer:
             # the Adadelta state dict is lazily created at the first eager
             # step, so the opt graph is only captured from step 2 on.
             try:
-                sg = _ShapeGraph(self, batch)  # capture ran fwd+bwd on batch
+                sg = _ShapeGraph(self, batch)
                 self.cache[key] = sg
+                sg.graph.replay()  # capture records only; compute this batch
                 loss = sg.loss
             except Exception as e:
                 import sys
