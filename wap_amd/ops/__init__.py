@@ -34,6 +34,16 @@ def get_backend() -> str:
     return _BACKEND
 
 
+_EAGER_OPS = set(
+    s for s in os.environ.get("WAP_EAGER_OPS", "").split(",") if s
+)
+
+
+def op_eager(name: str) -> bool:
+    """Per-op eager override (debug/bisect): WAP_EAGER_OPS=msc,attn,..."""
+    return name in _EAGER_OPS
+
+
 def use_hip(tensor) -> bool:
     """True if this op call should run the HIP kernel."""
     if _BACKEND == "eager":

@@ -24,7 +24,7 @@ from typing import Optional, Tuple
 import torch
 from torch import Tensor
 
-from . import use_hip
+from . import op_eager, use_hip
 
 def _up(t: Tensor) -> Tensor:
     """Upcast reduced-precision tensors to f32; leave f32/f64 untouched
@@ -43,7 +43,7 @@ class _AttentionScore(torch.autograd.Function):
         q_ = q.to(ua.dtype).contiguous()
         uf_ = uf.to(ua.dtype).contiguous()
         nu_ = _up(nu).contiguous()
-        if use_hip(ua):
+        if use_hip(ua) and not op_eager("attn"):
             from . import ext
             e, t = ext.require().attention_score_fwd(q_, ua.contiguous(), uf_, nu_)
         else:
@@ -58,7 +58,7 @@ class _AttentionScore(torch.autograd.Function):
         t, nu = ctx.saved_tensors
         qd, uad, ufd, nud = ctx.in_dtypes
         de = _up(de).contiguous()
-        if use_hip(t):
+        if use_hip(t) and not op_eager("attn"):
             from . import ext
             dq, dua, dnu = ext.require().attention_score_bwd(de, t, nu)
             return dq.to(qd), dua.to(uad), dua.clone().to(ufd), dnu.to(nud)
@@ -94,7 +94,7 @@ class _MaskedSoftmaxContext(torch.autograd.Function):
     def forward(ctx, e: Tensor, a: Tensor, mask: Tensor):
         e_ = _up(e).contiguous()
         mask_ = _up(mask).contiguous()
-        if use_hip(a):
+        if use_hip(a) and not op_eager("msc"):
             from . import ext
             alpha, ctxv = ext.require().masked_softmax_context_fwd(
                 e_, a.contiguous(), mask_
@@ -111,7 +111,7 @@ class _MaskedSoftmaxContext(torch.autograd.Function):
         ed, ad = ctx.in_dtypes
         g_alpha = _up(g_alpha).contiguous()
         g_ctx = _up(g_ctx).contiguous()
-        if use_hip(a):
+        if use_hip(a) and not op_eager("msc"):
             from . import ext
             de, da = ext.require().masked_softmax_context_bwd(g_alpha, g_ctx, alpha, a)
             return de.to(ed), da, None
@@ -145,7 +145,7 @@ class _Unfold1ch(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x: Tensor, kh: int, kw: int, stride: int, pad: int) -> Tensor:
         ctx.meta = (x.shape[-2], x.shape[-1], kh, kw, stride, pad, x.dtype)
-        if use_hip(x):
+        if use_hip(x) and not op_eager("unfold"):
             from . import ext
             b = x.shape[0]
             (cols,) = ext.require().im2col_1ch(
@@ -158,7 +158,7 @@ class _Unfold1ch(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dcols: Tensor):
         H, W, kh, kw, stride, pad, dt = ctx.meta
-        if use_hip(dcols):
+        if use_hip(dcols) and not op_eager("unfold"):
             from . import ext
             dx = ext.require().col2im_1ch(dcols.contiguous(), H, W, kh, kw,
                                           stride, pad)
@@ -205,7 +205,7 @@ class _GruGates(torch.autograd.Function):
         gx_ = gx.to(common).contiguous()
         gh_ = gh.to(common).contiguous()
         h_ = h.to(common).contiguous()
-        if use_hip(gx_):
+        if use_hip(gx_) and not op_eager("gru"):
             from . import ext
             h_new, cache = ext.require().gru_gates_fwd(gx_, gh_, h_)
             hip = True
