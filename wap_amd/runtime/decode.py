@@ -20,6 +20,15 @@ from __future__ import annotations
 from typing import Dict, List, Optional
 
 import torch
+
+
+def _capture_mode() -> str:
+    """hipGraph capture mode: thread_local allows other host threads (the
+    pinned-memory prefetcher) to touch HIP during capture; override with
+    WAP_CAPTURE_MODE=global for A/B."""
+    import os
+    return os.environ.get("WAP_CAPTURE_MODE", "thread_local")
+
 from torch import Tensor
 
 from ..data.vocab import EOS_ID
@@ -137,7 +146,7 @@ class BeamState:
         torch.cuda.synchronize()
         g = torch.cuda.CUDAGraph()
         ctx = torch.autocast("cuda", dtype=torch.bfloat16, enabled=autocast_bf16)
-        with ctx, torch.cuda.graph(g, capture_error_mode="thread_local"):
+        with ctx, torch.cuda.graph(g, capture_error_mode=_capture_mode()):
             _beam_step_(self.model, self.S, self.b, self.k, fused)
         self.graph = g
 

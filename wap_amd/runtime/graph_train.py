@@ -24,6 +24,15 @@ import torch
 import torch.distributed as dist
 
 
+def _capture_mode() -> str:
+    """hipGraph capture mode: thread_local allows other host threads (the
+    pinned-memory prefetcher) to touch HIP during capture; override with
+    WAP_CAPTURE_MODE=global for A/B."""
+    import os
+    return os.environ.get("WAP_CAPTURE_MODE", "thread_local")
+
+
+
 class GraphedTrainStep:
     """Capture-and-replay train step over static input buffers.
 
@@ -100,14 +109,14 @@ class GraphedTrainStep:
         torch.cuda.synchronize()
 
         self.g_fwdbwd = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.g_fwdbwd, capture_error_mode="thread_local"):
+        with torch.cuda.graph(self.g_fwdbwd, capture_error_mode=_capture_mode()):
             self.loss = self._fwd_bwd()
         # (Stream capture records without executing; grads in the buckets
         # are still the last warmup's.  The all-reduce keeps them finite and
         # rank-consistent for the opt-graph recording's tensor metadata.)
         self._allreduce()
         self.g_opt = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.g_opt, capture_error_mode="thread_local"):
+        with torch.cuda.graph(self.g_opt, capture_error_mode=_capture_mode()):
             self._clip_opt()
 
     def step(self, batch: Optional[Dict[str, torch.Tensor]] = None) -> torch.Tensor:
@@ -137,7 +146,7 @@ class _ShapeGraph:
         torch.cuda.current_stream().wait_stream(s)
         torch.cuda.synchronize()
         self.graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.graph, capture_error_mode="thread_local"):
+        with torch.cuda.graph(self.graph, capture_error_mode=_capture_mode()):
             self.loss = self._fwd_bwd()
         # NOTE: stream capture RECORDS the ops without executing them — the
         # caller must replay() to actually compute this batch.
@@ -204,7 +213,7 @@ class MultiShapeGraphedTrainer:
         # executes it once = this step's update).
         torch.cuda.synchronize()
         self.g_opt = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.g_opt, capture_error_mode="thread_local"):
+        with torch.cuda.graph(self.g_opt, capture_error_mode=_capture_mode()):
             if self.clip > 0:
                 torch.nn.utils.clip_grad_norm_(self.model.parameters(), self.clip)
             self.opt.step()
