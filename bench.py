@@ -118,12 +118,17 @@ def main() -> None:
         if on_gpu:
             torch.cuda.synchronize()
 
-    for _ in range(args.warmup):
+    import sys
+    for i in range(args.warmup):
         loss = train_step()
+        if rank == 0 and os.environ.get("WAP_LOSS_TRACE"):
+            print(f"warmup[{i}] loss={float(loss.item()):.4f}", file=sys.stderr)
     barrier_sync()
     t0 = time.perf_counter()
-    for _ in range(args.steps):
+    for i in range(args.steps):
         loss = train_step()
+        if rank == 0 and os.environ.get("WAP_LOSS_TRACE"):
+            print(f"step[{i}] loss={float(loss.item()):.4f}", file=sys.stderr)
     barrier_sync()
     elapsed = time.perf_counter() - t0
     final_loss = float(loss.item())
