@@ -70,7 +70,8 @@ class GraphedTrainStep:
         return out["loss"]
 
     def _clip_opt(self):
-        if self.clip > 0:
+        import os
+        if self.clip > 0 and os.environ.get("WAP_NO_CLIP", "0") != "1":
             torch.nn.utils.clip_grad_norm_(self.model.parameters(), self.clip)
         self.opt.step()
 
@@ -121,12 +122,16 @@ class GraphedTrainStep:
 
     def step(self, batch: Optional[Dict[str, torch.Tensor]] = None) -> torch.Tensor:
         """One training step; returns the (device) loss tensor."""
+        import os
         if batch is not None:
             for k in self.buf:
                 self.buf[k].copy_(batch[k], non_blocking=True)
         self.g_fwdbwd.replay()
         self._allreduce()
-        self.g_opt.replay()
+        if os.environ.get("WAP_OPT_GRAPH", "1") == "0":
+            self._clip_opt()   # eager optimizer (debug A/B)
+        else:
+            self.g_opt.replay()
         return self.loss
 
 
