@@ -109,29 +109,40 @@ class GraphedTrainStep:
         torch.cuda.current_stream().wait_stream(s)
         torch.cuda.synchronize()
 
-        self.g_fwdbwd = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.g_fwdbwd, capture_error_mode=_capture_mode()):
-            self.loss = self._fwd_bwd()
-        # (Stream capture records without executing; grads in the buckets
-        # are still the last warmup's.  The all-reduce keeps them finite and
-        # rank-consistent for the opt-graph recording's tensor metadata.)
-        self._allreduce()
-        self.g_opt = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(self.g_opt, capture_error_mode=_capture_mode()):
-            self._clip_opt()
+        # MEASURED HAZARD (ROCm 7.x): two back-to-back hipGraph replays on
+        # one stream can RACE (the second reads grads before the first's
+        # backward landed) -> exploding grads after a few steps.  For
+        # world==1 capture the WHOLE step as ONE graph; for world>1 keep
+        # two graphs with an explicit synchronize between the phases
+        # (step() below).
+        if self.ddp.world_size == 1:
+            self.g_all = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self.g_all, capture_error_mode=_capture_mode()):
+                self.loss = self._fwd_bwd()
+                self._clip_opt()
+        else:
+            self.g_all = None
+            self.g_fwdbwd = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self.g_fwdbwd,
+                                  capture_error_mode=_capture_mode()):
+                self.loss = self._fwd_bwd()
+            self._allreduce()
+            self.g_opt = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(self.g_opt, capture_error_mode=_capture_mode()):
+                self._clip_opt()
 
     def step(self, batch: Optional[Dict[str, torch.Tensor]] = None) -> torch.Tensor:
         """One training step; returns the (device) loss tensor."""
-        import os
         if batch is not None:
             for k in self.buf:
                 self.buf[k].copy_(batch[k], non_blocking=True)
+        if self.g_all is not None:
+            self.g_all.replay()
+            return self.loss
         self.g_fwdbwd.replay()
+        torch.cuda.synchronize()  # replay-replay race guard (see initialize)
         self._allreduce()
-        if os.environ.get("WAP_OPT_GRAPH", "1") == "0":
-            self._clip_opt()   # eager optimizer (debug A/B)
-        else:
-            self.g_opt.replay()
+        self.g_opt.replay()
         return self.loss
 
 
@@ -153,6 +164,13 @@ class _ShapeGraph:
         self.graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.graph, capture_error_mode=_capture_mode()):
             self.loss = self._fwd_bwd()
+            if trainer.ddp.world_size == 1:
+                # single-rank: fold clip+opt into the same graph (avoids the
+                # replay-replay race) once optimizer state exists
+                trainer._clip_opt_ops()
+                self.has_opt = True
+            else:
+                self.has_opt = False
         # NOTE: stream capture RECORDS the ops without executing them — the
         # caller must replay() to actually compute this batch.
 
@@ -210,6 +228,11 @@ class MultiShapeGraphedTrainer:
                 h.wait()
                 b.buffer.div_(self.ddp.world_size)
 
+    def _clip_opt_ops(self):
+        if self.clip > 0:
+            torch.nn.utils.clip_grad_norm_(self.model.parameters(), self.clip)
+        self.opt.step()
+
     def _opt_step(self):
         if self.g_opt is not None:
             self.g_opt.replay()
@@ -234,12 +257,21 @@ class MultiShapeGraphedTrainer:
         return out["loss"]
 
     def step(self, batch: Dict[str, torch.Tensor]) -> torch.Tensor:
+        if not self.opt.state:
+            # very first step: all-eager (initializes optimizer state so
+            # every captured shape graph can include clip+opt at world==1)
+            loss = self._eager_fwd_bwd(batch)
+            self._allreduce()
+            self._clip_opt_ops()
+            return loss
         key = tuple(batch["x"].shape) + tuple(batch["y"].shape)
         sg = self.cache.get(key)
         if sg is not None:
             for k in sg.buf:
                 sg.buf[k].copy_(batch[k], non_blocking=True)
             sg.graph.replay()
+            if sg.has_opt:
+                return sg.loss
             loss = sg.loss
         elif len(self.cache) < self.max_graphs and key not in self._failed:
             # Optimizer state must exist before its graph can be captured;
@@ -249,6 +281,8 @@ class MultiShapeGraphedTrainer:
                 sg = _ShapeGraph(self, batch)
                 self.cache[key] = sg
                 sg.graph.replay()  # capture records only; compute this batch
+                if sg.has_opt:
+                    return sg.loss
                 loss = sg.loss
             except Exception as e:
                 import sys
@@ -259,12 +293,8 @@ class MultiShapeGraphedTrainer:
         else:
             self.n_eager_fallback += 1
             loss = self._eager_fwd_bwd(batch)
+        # split path (world>1, or a shape captured without opt):
+        torch.cuda.synchronize()  # replay-replay race guard
         self._allreduce()
-        if self.g_opt is None and not self.opt.state:
-            # very first step: eager optimizer step initializes state
-            if self.clip > 0:
-                torch.nn.utils.clip_grad_norm_(self.model.parameters(), self.clip)
-            self.opt.step()
-        else:
-            self._opt_step()
+        self._opt_step()
         return loss
