@@ -87,6 +87,16 @@ DEV float block_max(float v, float* red) {
 
 DEV float sigmoidf(float x) { return 1.f / (1.f + __expf(-x)); }
 
+// Explicit zero-fill kernel for atomic-accumulator buffers.  torch::zeros
+// lowers to an async memset, which is NOT reliably recorded into captured
+// hipGraphs on some ROCm driver builds (measured: atomically-accumulated
+// grads grow across graph replays on a subset of boxes).  A plain kernel
+// is always captured.
+__global__ void zero_f32_kernel(float* __restrict__ p, long n) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) p[i] = 0.f;
+}
+
 // ------------------------------------------- attention score forward
 // e[b,l] = sum_k nu[k] * tanh(q[b,k] + ua[b,l,k] + uf[b,l,k])
 // One wave per (b,l); q and nu staged in LDS per block.
@@ -502,9 +512,14 @@ std::vector<torch::Tensor> attention_score_bwd(
   CHECK_GPU(de); CHECK_GPU(t); CHECK_GPU(nu);
   const int B = t.size(0), L = t.size(1), A = t.size(2);
   TORCH_CHECK(A <= 4 * 256, "A too large for attn_score_bwd accumulators");
-  auto dq = torch::zeros({B, A}, t.options().dtype(at::kFloat));
+  auto dq = torch::empty({B, A}, t.options().dtype(at::kFloat));
   auto dua = torch::empty_like(t);
-  auto dnu = torch::zeros({A}, t.options().dtype(at::kFloat));
+  auto dnu = torch::empty({A}, t.options().dtype(at::kFloat));
+  const long zn = (long)B * A;
+  hipLaunchKernelGGL(zero_f32_kernel, dim3((zn + 255) / 256), dim3(256), 0,
+                     cur_stream(), ptr<float>(dq), zn);
+  hipLaunchKernelGGL(zero_f32_kernel, dim3((A + 255) / 256), dim3(256), 0,
+                     cur_stream(), ptr<float>(dnu), (long)A);
   const dim3 grid((L + ATTN_BWD_LCH - 1) / ATTN_BWD_LCH, B);
   if (is_bf16(t)) {
     hipLaunchKernelGGL(attn_score_bwd_kernel<__hip_bfloat16>, grid, dim3(256),
@@ -568,7 +583,9 @@ std::vector<torch::Tensor> masked_softmax_context_bwd(
   const int B = a.size(0), L = a.size(1), D = a.size(2);
   auto de = torch::empty({B, L}, g_alpha.options().dtype(at::kFloat));
   auto da = torch::empty_like(a);
-  auto sscr = torch::zeros({B}, g_alpha.options().dtype(at::kFloat));
+  auto sscr = torch::empty({B}, g_alpha.options().dtype(at::kFloat));
+  hipLaunchKernelGGL(zero_f32_kernel, dim3(1), dim3(256), 0, cur_stream(),
+                     ptr<float>(sscr), (long)B);
   const int smem = (16 + D) * sizeof(float);
   const dim3 grid((L + MSC_BWD_LCH - 1) / MSC_BWD_LCH, B);
   if (is_bf16(a)) {
