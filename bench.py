@@ -147,7 +147,7 @@ def main() -> None:
     def barrier_sync():
         if world > 1:
             torch.distributed.barrier()
-        if on_gpu:
+        if on_gpu and not os.environ.get("WAP_NO_BARRIER"):
             torch.cuda.synchronize()
 
     import sys
@@ -159,6 +159,8 @@ def main() -> None:
     t0 = time.perf_counter()
     for i in range(args.steps):
         loss = train_step()
+        if os.environ.get("WAP_SYNC_EVERY"):
+            torch.cuda.synchronize()
         if rank == 0 and os.environ.get("WAP_LOSS_TRACE"):
             print(f"step[{i}] loss={float(loss.item()):.4f}", file=sys.stderr)
     barrier_sync()
