@@ -145,10 +145,15 @@ def main() -> None:
         return out["loss"]
 
     def barrier_sync():
+        # STREAM-scope sync, never torch.cuda.synchronize(): a device-wide
+        # synchronize between hipGraph replays corrupts the next replay's
+        # backward on ROCm 7.x (measured: NaN at the step after every
+        # device sync; stream syncs are safe).  All work is on the current
+        # stream, so stream sync is sufficient for timing.
         if world > 1:
             torch.distributed.barrier()
         if on_gpu and not os.environ.get("WAP_NO_BARRIER"):
-            torch.cuda.synchronize()
+            torch.cuda.current_stream().synchronize()
 
     import sys
     for i in range(args.warmup):
@@ -159,8 +164,6 @@ def main() -> None:
     t0 = time.perf_counter()
     for i in range(args.steps):
         loss = train_step()
-        if os.environ.get("WAP_SYNC_EVERY"):
-            torch.cuda.synchronize()
         if rank == 0 and os.environ.get("WAP_LOSS_TRACE"):
             print(f"step[{i}] loss={float(loss.item()):.4f}", file=sys.stderr)
     barrier_sync()
@@ -184,7 +187,7 @@ def main() -> None:
         with autocast():
             beam_decode(model, x[:db], x_mask[:db], beam=args.beam, maxlen=32)
         if on_gpu:
-            torch.cuda.synchronize()
+            torch.cuda.current_stream().synchronize()
         tok = 0
         td0 = time.perf_counter()
         for _ in range(decode_rounds):
@@ -193,7 +196,7 @@ def main() -> None:
                                    beam=args.beam, maxlen=64)
             tok += sum(max(len(h), 1) for h in hyps)
         if on_gpu:
-            torch.cuda.synchronize()
+            torch.cuda.current_stream().synchronize()
         td = time.perf_counter() - td0
         extra["decode_tok_s"] = tok / td
         extra["decode_beam"] = args.beam

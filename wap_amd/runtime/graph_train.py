@@ -140,7 +140,9 @@ class GraphedTrainStep:
             self.g_all.replay()
             return self.loss
         self.g_fwdbwd.replay()
-        torch.cuda.synchronize()  # replay-replay race guard (see initialize)
+        # stream-scope only: device-wide sync between replays corrupts the
+        # next replay's backward (ROCm 7.x, measured)
+        torch.cuda.current_stream().synchronize()
         self._allreduce()
         self.g_opt.replay()
         return self.loss
@@ -294,7 +296,7 @@ class MultiShapeGraphedTrainer:
             self.n_eager_fallback += 1
             loss = self._eager_fwd_bwd(batch)
         # split path (world>1, or a shape captured without opt):
-        torch.cuda.synchronize()  # replay-replay race guard
+        torch.cuda.current_stream().synchronize()  # stream-scope only
         self._allreduce()
         self._opt_step()
         return loss
