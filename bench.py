@@ -145,15 +145,17 @@ def main() -> None:
         return out["loss"]
 
     def barrier_sync():
-        # STREAM-scope sync, never torch.cuda.synchronize(): a device-wide
-        # synchronize between hipGraph replays corrupts the next replay's
-        # backward on ROCm 7.x (measured: NaN at the step after every
-        # device sync; stream syncs are safe).  All work is on the current
-        # stream, so stream sync is sufficient for timing.
+        # Sync via a D2H scalar read (hipMemcpyWithStream), NEVER
+        # hipStreamSynchronize / hipDeviceSynchronize: on ROCm 7.x both of
+        # those corrupt the next hipGraph replay's backward (measured:
+        # grads explode exactly one step after every such sync; the
+        # memcpy-style sync .item() performs is safe).  The read drains the
+        # current stream, which carries all our work, so timing is exact.
         if world > 1:
             torch.distributed.barrier()
         if on_gpu and not os.environ.get("WAP_NO_BARRIER"):
-            torch.cuda.current_stream().synchronize()
+            from wap_amd.runtime.graph_train import safe_stream_sync
+            safe_stream_sync()
 
     import sys
     for i in range(args.warmup):
@@ -187,7 +189,8 @@ def main() -> None:
         with autocast():
             beam_decode(model, x[:db], x_mask[:db], beam=args.beam, maxlen=32)
         if on_gpu:
-            torch.cuda.current_stream().synchronize()
+            from wap_amd.runtime.graph_train import safe_stream_sync
+            safe_stream_sync()
         tok = 0
         td0 = time.perf_counter()
         for _ in range(decode_rounds):
@@ -196,7 +199,8 @@ def main() -> None:
                                    beam=args.beam, maxlen=64)
             tok += sum(max(len(h), 1) for h in hyps)
         if on_gpu:
-            torch.cuda.current_stream().synchronize()
+            from wap_amd.runtime.graph_train import safe_stream_sync
+            safe_stream_sync()
         td = time.perf_counter() - td0
         extra["decode_tok_s"] = tok / td
         extra["decode_beam"] = args.beam

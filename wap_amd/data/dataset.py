@@ -227,8 +227,9 @@ class BucketLoader:
     def __iter__(self) -> Iterator[Dict[str, torch.Tensor]]:
         order = list(range(len(self.local)))
         random.Random(self.seed * 1_000_003 + self.epoch).shuffle(order)
+        import os
         dev = torch.device(self.device)
-        if dev.type != "cuda":
+        if dev.type != "cuda" or os.environ.get("WAP_NO_PREFETCH"):
             for i in order:
                 yield prepare_batch(self.local[i], self.device, self.dtype,
                                     pad_to=self.pad_to)

@@ -24,6 +24,21 @@ import torch
 import torch.distributed as dist
 
 
+_SYNC_SCRATCH = None
+
+
+def safe_stream_sync() -> None:
+    """Drain the current stream WITHOUT hipStreamSynchronize /
+    hipDeviceSynchronize: on ROCm 7.x those corrupt the next hipGraph
+    replay's backward (measured, deterministic on most boxes).  A D2H
+    scalar read (hipMemcpyWithStream under .item()) drains the stream and
+    is measured safe."""
+    global _SYNC_SCRATCH
+    if _SYNC_SCRATCH is None or not _SYNC_SCRATCH.is_cuda:
+        _SYNC_SCRATCH = torch.zeros(1, device="cuda")
+    _SYNC_SCRATCH.item()
+
+
 def _capture_mode() -> str:
     """hipGraph capture mode: thread_local allows other host threads (the
     pinned-memory prefetcher) to touch HIP during capture; override with
@@ -140,9 +155,7 @@ class GraphedTrainStep:
             self.g_all.replay()
             return self.loss
         self.g_fwdbwd.replay()
-        # stream-scope only: device-wide sync between replays corrupts the
-        # next replay's backward (ROCm 7.x, measured)
-        torch.cuda.current_stream().synchronize()
+        safe_stream_sync()  # see docstring: never hipStreamSynchronize here
         self._allreduce()
         self.g_opt.replay()
         return self.loss
@@ -296,7 +309,7 @@ class MultiShapeGraphedTrainer:
             self.n_eager_fallback += 1
             loss = self._eager_fwd_bwd(batch)
         # split path (world>1, or a shape captured without opt):
-        torch.cuda.current_stream().synchronize()  # stream-scope only
+        safe_stream_sync()
         self._allreduce()
         self._opt_step()
         return loss
