@@ -136,14 +136,15 @@ class BeamState:
     def capture(self, fused: bool = True, autocast_bf16: bool = False) -> None:
         """Capture one beam step into a hipGraph (CUDAGraph on PyTorch-ROCm);
         subsequent ``replay()`` calls re-run it on the static buffers."""
-        torch.cuda.synchronize()
+        from .graph_train import safe_stream_sync
+        safe_stream_sync()  # NEVER hipDeviceSynchronize after graph replays
         side = torch.cuda.Stream()
         side.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(side):  # warm up allocator/MIOpen off-graph
             for _ in range(2):
                 _beam_step_(self.model, self.S, self.b, self.k, fused)
         torch.cuda.current_stream().wait_stream(side)
-        torch.cuda.synchronize()
+        safe_stream_sync()
         g = torch.cuda.CUDAGraph()
         ctx = torch.autocast("cuda", dtype=torch.bfloat16, enabled=autocast_bf16)
         with ctx, torch.cuda.graph(g, capture_error_mode=_capture_mode()):

@@ -122,7 +122,7 @@ class GraphedTrainStep:
                 self._allreduce()
                 self._clip_opt()
         torch.cuda.current_stream().wait_stream(s)
-        torch.cuda.synchronize()
+        safe_stream_sync()
 
         # MEASURED HAZARD (ROCm 7.x): two back-to-back hipGraph replays on
         # one stream can RACE (the second reads grads before the first's
@@ -175,7 +175,7 @@ class _ShapeGraph:
         with torch.cuda.stream(s):
             self._fwd_bwd()
         torch.cuda.current_stream().wait_stream(s)
-        torch.cuda.synchronize()
+        safe_stream_sync()
         self.graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.graph, capture_error_mode=_capture_mode()):
             self.loss = self._fwd_bwd()
@@ -252,9 +252,8 @@ class MultiShapeGraphedTrainer:
         if self.g_opt is not None:
             self.g_opt.replay()
             return
-        # First step on every rank: capture the clip+opt graph (capture
-        # executes it once = this step's update).
-        torch.cuda.synchronize()
+        # First step on every rank: capture the clip+opt graph.
+        safe_stream_sync()
         self.g_opt = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.g_opt, capture_error_mode=_capture_mode()):
             if self.clip > 0:
