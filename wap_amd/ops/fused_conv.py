@@ -83,13 +83,15 @@ class _BNReluConv(torch.autograd.Function):
 
 def can_fuse(x: Tensor, weight: Tensor) -> bool:
     import os
-    mode = os.environ.get("WAP_FUSED_CONV", "1x1")
+    # Default OFF: the kernel itself beats MIOpen 2-3x on the 1x1 shapes
+    # (tools/bench_conv.py) but the training wrapper's batch-stat and
+    # f32-recompute overhead currently costs more than the kernel saves
+    # (profiles/: var_mean + casts ~16% of the step).  Opt in with
+    # WAP_FUSED_CONV=1x1 (1x1 convs) or =all.
+    mode = os.environ.get("WAP_FUSED_CONV", "0")
     if mode == "0":
         return False
     if mode == "1x1" and weight.shape[2] != 1:
-        # measured (tools/bench_conv.py): the MFMA kernel beats MIOpen
-        # 2-3x on the 1x1 bottleneck convs, but the single-buffered 3x3
-        # path still trails igemm -- 3x3 stays on MIOpen until tuned.
         return False
     return (
         use_hip(x)
