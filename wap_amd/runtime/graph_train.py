@@ -170,10 +170,13 @@ class _ShapeGraph:
         self.buf = {k: batch[k].clone() for k in ("x", "x_mask", "y", "y_mask")}
         # Warmup fwd+bwd only — NO collectives, NO optimizer: lazy capture
         # must not disturb the per-step all-reduce alignment across ranks.
+        # 3 iterations: backward capture needs the autograd engine fully
+        # warmed (fewer warmups segfaulted capture_end on MI355X).
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
-            self._fwd_bwd()
+            for _ in range(3):
+                self._fwd_bwd()
         torch.cuda.current_stream().wait_stream(s)
         safe_stream_sync()
         self.graph = torch.cuda.CUDAGraph()
