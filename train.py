@@ -77,7 +77,9 @@ def load_split(cfg: WAPConfig, vocab: Vocab, split: str, seed: int):
 def validate(model, buckets, vocab, cfg, device, autocast_ctx) -> float:
     """Greedy-decode the validation buckets -> WER (rank 0 only)."""
     pairs = []
-    loader = BucketLoader(buckets, device=device)
+    pad_to = ((cfg.graph_pad_hw, cfg.graph_pad_hw, cfg.graph_pad_t, 1)
+              if torch.device(device).type == "cuda" else None)
+    loader = BucketLoader(buckets, device=device, pad_to=pad_to)
     with torch.no_grad():
         for batch in loader:
             with autocast_ctx():
@@ -89,7 +91,8 @@ def validate(model, buckets, vocab, cfg, device, autocast_ctx) -> float:
                 )
             refs = batch["y"]
             masks = batch["y_mask"]
-            for i, hyp in enumerate(hyps):
+            for i in range(len(batch["uids"])):   # skip batch-padded rows
+                hyp = hyps[i]
                 t = int(masks[i].sum().item()) - 1  # exclude eos slot
                 ref_ids = refs[i, :t].tolist()
                 pairs.append((
@@ -139,14 +142,18 @@ def main() -> None:
     use_graph = device.type == "cuda" and cfg.graph != "off"
     trainer = None
     pad_to = None
+    if device.type == "cuda":
+        # Collapse bucket shapes onto a grid: (a) lets captured graphs be
+        # reused, (b) works around a MIOpen BatchNorm segfault after a
+        # handful of distinct input shapes (ROCm 7.x, reproduced with all
+        # WAP kernels disabled; see PERF.md).
+        pad_to = (cfg.graph_pad_hw, cfg.graph_pad_hw, cfg.graph_pad_t,
+                  cfg.batch_size)
     if use_graph:
         from wap_amd.runtime.graph_train import MultiShapeGraphedTrainer
         trainer = MultiShapeGraphedTrainer(
             model, ddp, optimizer, clip=cfg.grad_clip, autocast_bf16=use_bf16
         )
-        # collapse bucket shapes onto a grid so captured graphs are reused
-        pad_to = (cfg.graph_pad_hw, cfg.graph_pad_hw, cfg.graph_pad_t,
-                  cfg.batch_size)
     loader = BucketLoader(
         train_buckets, rank=rank, world_size=world, seed=cfg.seed,
         device=device, pad_to=pad_to,
