@@ -59,8 +59,9 @@ def main() -> None:
     rank, world = info["rank"], info["world_size"]
     on_gpu = torch.cuda.is_available()
     if on_gpu:
-        torch.cuda.set_device(info["local_rank"])
-        device = torch.device("cuda", info["local_rank"])
+        dev_idx = info["local_rank"] % torch.cuda.device_count()
+        torch.cuda.set_device(dev_idx)
+        device = torch.device("cuda", dev_idx)
     else:
         device = torch.device("cpu")
     torch.manual_seed(1234 + rank)

@@ -186,3 +186,33 @@ def test_channels_last_matches_contiguous():
         a2, _ = m.encode(x, xm)
     err = (a1 - a2).abs().max().item()
     assert err < 1e-3, err
+
+
+def test_graphed_steps_stay_finite_across_syncs():
+    """Regression test for the ROCm device-sync/graph-replay corruption:
+    replayed steps interleaved with safe_stream_sync must stay finite."""
+    from wap_amd import ops
+    from wap_amd.models import WAPModel
+    from wap_amd.parallel import BucketedDDP
+    from wap_amd.runtime.graph_train import GraphedTrainStep, safe_stream_sync
+    ops.set_backend("auto")
+    dev = torch.device("cuda", 0)
+    torch.manual_seed(33)
+    m = WAPModel(n_tokens=40, encoder="densenet",
+                 encoder_kw={"growth": 8, "layers_per_block": 2}).to(dev)
+    opt = torch.optim.Adadelta(m.parameters(), lr=1.0, weight_decay=1e-4)
+    ddp = BucketedDDP(m)
+    batch = {
+        "x": torch.rand(2, 1, 64, 128, device=dev),
+        "x_mask": torch.ones(2, 64, 128, device=dev),
+        "y": torch.randint(1, 40, (2, 6), device=dev),
+        "y_mask": torch.ones(2, 6, device=dev),
+    }
+    g = GraphedTrainStep(m, ddp, opt, clip=100.0, autocast_bf16=True)
+    g.initialize(batch)
+    losses = []
+    for i in range(8):
+        loss = g.step()
+        safe_stream_sync()
+        losses.append(float(loss.item()))
+    assert all(l == l and l < 100 for l in losses), losses

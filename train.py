@@ -40,8 +40,9 @@ from wap_amd.runtime import (
 def pick_device(cfg: WAPConfig, local_rank: int) -> torch.device:
     if cfg.device == "cpu" or (cfg.device == "auto" and not torch.cuda.is_available()):
         return torch.device("cpu")
-    torch.cuda.set_device(local_rank)
-    return torch.device("cuda", local_rank)
+    idx = local_rank % torch.cuda.device_count()
+    torch.cuda.set_device(idx)
+    return torch.device("cuda", idx)
 
 
 def build_optimizer(cfg: WAPConfig, model: torch.nn.Module) -> torch.optim.Optimizer:

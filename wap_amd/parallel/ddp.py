@@ -151,6 +151,8 @@ def init_distributed(device: str = "auto") -> Dict[str, int]:
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     if world > 1 and not dist.is_initialized():
-        backend = "nccl" if (device != "cpu" and torch.cuda.is_available()) else "gloo"
+        backend = os.environ.get("WAP_DIST_BACKEND") or (
+            "nccl" if (device != "cpu" and torch.cuda.is_available()) else "gloo"
+        )
         dist.init_process_group(backend=backend)
     return {"rank": rank, "world_size": world, "local_rank": local_rank}
