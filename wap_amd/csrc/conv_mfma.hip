@@ -206,3 +206,158 @@ void bnrelu_conv_mfma_fwd_bk96(
   bnrelu_conv_mfma_body<96>(x, w, scale, shift, out, Bn, H, W, C, KOUT,
                             R, S, relu_on);
 }
+
+// ---------------------------------------------------------------------
+// FP8 (OCP e4m3) variant — BASELINE config 5.  x and w are pre-quantized
+// e4m3 bytes; their per-tensor dequant scales are folded into the BN
+// affine (x) and the output scale (w), so the kernel math is:
+//   out[p,k] = out_scale * sum relu8(x_q[p+D,c]*scale_c + shift_c) * w_q[k,..]
+// where relu8 re-quantizes the normalized activation to e4m3 for the MFMA
+// (v_mfma_f32_16x16x32_fp8_fp8, f32 accumulation).  Same tiling as bf16.
+// ---------------------------------------------------------------------
+
+#include <hip/hip_fp8.h>
+
+typedef __hip_fp8_e4m3 fp8_t;
+typedef long i64frag;
+
+DEVC float f8tof(fp8_t v) { return (float)v; }
+DEVC fp8_t ftof8(float v) { return fp8_t(v); }
+
+template <int BKC>
+__device__ __forceinline__
+void bnrelu_conv_mfma_fp8_body(
+    const fp8_t* __restrict__ x, const fp8_t* __restrict__ w,
+    const float* __restrict__ scale, const float* __restrict__ shift,
+    float out_scale, __hip_bfloat16* __restrict__ out,
+    int Bn, int H, int W, int C, int KOUT, int R, int S, int relu_on) {
+  __shared__ fp8_t lA[BM][BKC];
+  __shared__ fp8_t lB[96][BKC];
+  __shared__ float lscale[BKC], lshift[BKC];
+
+  const int HW = H * W;
+  const long M = (long)Bn * HW;
+  const long m0 = (long)blockIdx.x * BM;
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wv = tid >> 6;
+  const int PR = R / 2, PS = S / 2;
+  const int NFRAG = (KOUT + 15) / 16;
+  f32x4 acc[2][6];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 6; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  for (int r = 0; r < R; ++r) {
+    for (int s = 0; s < S; ++s) {
+      for (int c0 = 0; c0 < C; c0 += BKC) {
+        if (tid < BKC) {
+          const int c = c0 + tid;
+          lscale[tid] = c < C ? scale[c] : 0.f;
+          lshift[tid] = c < C ? shift[c] : 0.f;
+        }
+        __syncthreads();
+        constexpr int PIECES_PER_ROW = BKC / 8;   // 8 fp8 per piece
+        constexpr int A_PIECES = BM * PIECES_PER_ROW;
+#pragma unroll
+        for (int i = 0; i < A_PIECES / 256; ++i) {
+          const int flat = tid * (A_PIECES / 256) + i;
+          const int arow = flat / PIECES_PER_ROW;
+          const int apc = flat % PIECES_PER_ROW;
+          const long p = m0 + arow;
+          bool inb = false;
+          long base = 0;
+          if (p < M) {
+            const int b = (int)(p / HW);
+            const int rem = (int)(p % HW);
+            const int y = rem / W + r - PR;
+            const int xw = rem % W + s - PS;
+            if (y >= 0 && y < H && xw >= 0 && xw < W) {
+              inb = true;
+              base = (((long)b * H + y) * W + xw) * C + c0 + apc * 8;
+            }
+          }
+#pragma unroll
+          for (int e = 0; e < 8; ++e) {
+            const int cc = apc * 8 + e;
+            const int c = c0 + cc;
+            float v = 0.f;
+            if (inb && c < C) {
+              v = f8tof(x[base + e]) * lscale[cc] + lshift[cc];
+              if (relu_on) v = fmaxf(v, 0.f);
+            }
+            lA[arow][cc] = ftof8(v);
+          }
+        }
+        for (int flat = tid; flat < NFRAG * 16 * PIECES_PER_ROW;
+             flat += blockDim.x) {
+          const int k = flat / PIECES_PER_ROW;
+          const int bpc = flat % PIECES_PER_ROW;
+          if (k < 96) {
+#pragma unroll
+            for (int e = 0; e < 8; ++e) {
+              const int c = c0 + bpc * 8 + e;
+              lB[k][bpc * 8 + e] = (k < KOUT && c < C)
+                  ? w[(((long)k * R + r) * S + s) * C + c] : ftof8(0.f);
+            }
+          }
+        }
+        __syncthreads();
+#pragma unroll
+        for (int ks = 0; ks < BKC / 32; ++ks) {
+#pragma unroll
+          for (int mf = 0; mf < 2; ++mf) {
+            const int arow = wv * 32 + mf * 16 + (lane & 15);
+            const int kk = ks * 32 + (lane >> 4) * 8;
+            const i64frag afrag = *(const i64frag*)&lA[arow][kk];
+#pragma unroll
+            for (int nf = 0; nf < 6; ++nf) {
+              if (nf >= NFRAG) break;
+              const int brow = nf * 16 + (lane & 15);
+              const i64frag bfrag = *(const i64frag*)&lB[brow][kk];
+              acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+                  afrag, bfrag, acc[mf][nf], 0, 0, 0);
+            }
+          }
+        }
+        __syncthreads();
+      }
+    }
+  }
+#pragma unroll
+  for (int mf = 0; mf < 2; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 6; ++nf) {
+      if (nf >= NFRAG) break;
+      const int k = nf * 16 + (lane & 15);
+      if (k >= KOUT) continue;
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int mrow = wv * 32 + mf * 16 + (lane >> 4) * 4 + reg;
+        const long p = m0 + mrow;
+        if (p < M) out[p * KOUT + k] = f2bf(acc[mf][nf][reg] * out_scale);
+      }
+    }
+  }
+}
+
+extern "C" __global__ __launch_bounds__(256, 2)
+void bnrelu_conv_mfma_fp8_fwd(
+    const fp8_t* __restrict__ x, const fp8_t* __restrict__ w,
+    const float* __restrict__ scale, const float* __restrict__ shift,
+    float out_scale, __hip_bfloat16* __restrict__ out,
+    int Bn, int H, int W, int C, int KOUT, int R, int S, int relu_on) {
+  bnrelu_conv_mfma_fp8_body<32>(x, w, scale, shift, out_scale, out,
+                                Bn, H, W, C, KOUT, R, S, relu_on);
+}
+
+extern "C" __global__ __launch_bounds__(256, 2)
+void bnrelu_conv_mfma_fp8_fwd_bk96(
+    const fp8_t* __restrict__ x, const fp8_t* __restrict__ w,
+    const float* __restrict__ scale, const float* __restrict__ shift,
+    float out_scale, __hip_bfloat16* __restrict__ out,
+    int Bn, int H, int W, int C, int KOUT, int R, int S, int relu_on) {
+  bnrelu_conv_mfma_fp8_body<96>(x, w, scale, shift, out_scale, out,
+                                Bn, H, W, C, KOUT, R, S, relu_on);
+}

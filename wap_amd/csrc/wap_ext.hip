@@ -23,6 +23,7 @@
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
+#include <hip/hip_fp8.h>
 
 #include <vector>
 
@@ -747,6 +748,12 @@ extern "C" __global__ void bnrelu_conv_mfma_fwd(
 extern "C" __global__ void bnrelu_conv_mfma_fwd_bk96(
     const __hip_bfloat16*, const __hip_bfloat16*, const float*, const float*,
     __hip_bfloat16*, int, int, int, int, int, int, int, int);
+extern "C" __global__ void bnrelu_conv_mfma_fp8_fwd(
+    const __hip_fp8_e4m3*, const __hip_fp8_e4m3*, const float*, const float*,
+    float, __hip_bfloat16*, int, int, int, int, int, int, int, int);
+extern "C" __global__ void bnrelu_conv_mfma_fp8_fwd_bk96(
+    const __hip_fp8_e4m3*, const __hip_fp8_e4m3*, const float*, const float*,
+    float, __hip_bfloat16*, int, int, int, int, int, int, int, int);
 
 torch::Tensor bnrelu_conv_fwd(
     torch::Tensor x, torch::Tensor w, torch::Tensor scale,
@@ -777,8 +784,39 @@ torch::Tensor bnrelu_conv_fwd(
   return out;
 }
 
+torch::Tensor bnrelu_conv_fwd_fp8(
+    torch::Tensor x, torch::Tensor w, torch::Tensor scale,
+    torch::Tensor shift, double out_scale, bool relu_on) {
+  TORCH_CHECK(x.is_cuda() && w.is_cuda());
+  TORCH_CHECK(x.scalar_type() == at::kFloat8_e4m3fn &&
+              w.scalar_type() == at::kFloat8_e4m3fn,
+              "bnrelu_conv_fp8: float8_e4m3fn only");
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast) &&
+              w.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "x/w must be channels_last");
+  const int B = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  const int K = w.size(0), R = w.size(2), S = w.size(3);
+  TORCH_CHECK(w.size(1) == C && K <= 96 && (R % 2) && (S % 2));
+  auto out = at::empty({B, K, H, W},
+                       x.options().dtype(at::kBFloat16),
+                       at::MemoryFormat::ChannelsLast);
+  const long M = (long)B * H * W;
+  const dim3 grid((M + 127) / 128);
+  auto kern = (C % 96 == 0) ? bnrelu_conv_mfma_fp8_fwd_bk96
+                            : bnrelu_conv_mfma_fp8_fwd;
+  hipLaunchKernelGGL(kern, grid, dim3(256), 0, cur_stream(),
+                     reinterpret_cast<const __hip_fp8_e4m3*>(x.data_ptr()),
+                     reinterpret_cast<const __hip_fp8_e4m3*>(w.data_ptr()),
+                     cptr<float>(scale), cptr<float>(shift),
+                     (float)out_scale, ptr<__hip_bfloat16>(out),
+                     B, H, W, C, K, R, S, relu_on ? 1 : 0);
+  HIP_OK();
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bnrelu_conv_fwd", &bnrelu_conv_fwd);
+  m.def("bnrelu_conv_fwd_fp8", &bnrelu_conv_fwd_fp8);
   m.def("attention_score_fwd", &attention_score_fwd);
   m.def("attention_score_bwd", &attention_score_bwd);
   m.def("masked_softmax_context_fwd", &masked_softmax_context_fwd);

@@ -110,3 +110,28 @@ def bnrelu_conv(x: Tensor, weight: Tensor, bn: torch.nn.BatchNorm2d) -> Tensor:
     w = weight.to(torch.bfloat16).contiguous(memory_format=torch.channels_last)
     return _BNReluConv.apply(x, w, bn.weight, bn.bias, mean, var,
                              bn.eps, bn.training)
+
+
+# ----------------------------------------------------------------- fp8 path
+# BASELINE config 5: fp8 (OCP e4m3) MFMA conv.  Per-tensor amax scaling;
+# the activation dequant scale folds into the BN affine, the weight scale
+# into the output epilogue, so quantization adds no extra kernel passes.
+
+def quantize_e4m3(t: Tensor):
+    """-> (float8_e4m3fn tensor (channels_last for 4-D), f32 scale)."""
+    amax = t.abs().amax().float().clamp_min(1e-12)
+    s = amax / 448.0
+    q = (t.float() / s).clamp(-448.0, 448.0).to(torch.float8_e4m3fn)
+    if q.dim() == 4:
+        q = q.contiguous(memory_format=torch.channels_last)
+    return q, s
+
+
+@torch.no_grad()
+def bnrelu_conv_fp8(x_q: Tensor, sx: Tensor, w_q: Tensor, sw: Tensor,
+                    scale: Tensor, shift: Tensor, relu: bool = True) -> Tensor:
+    """relu(bn(dequant(x_q))) -> conv with e4m3 MFMA; returns bf16 NHWC."""
+    from . import ext
+    return ext.require().bnrelu_conv_fwd_fp8(
+        x_q, w_q, (scale.float() * sx).contiguous(),
+        shift.float().contiguous(), float(sw), relu)
