@@ -275,9 +275,18 @@ class MultiShapeGraphedTrainer:
 
     def step(self, batch: Dict[str, torch.Tensor]) -> torch.Tensor:
         if not self.opt.state:
-            # very first step: all-eager (initializes optimizer state so
-            # every captured shape graph can include clip+opt at world==1)
-            loss = self._eager_fwd_bwd(batch)
+            # Very first step: all-eager (initializes optimizer state so
+            # every captured shape graph can include clip+opt at world==1).
+            # Run it on a SIDE stream: the process's first backward creates
+            # the per-param AccumulateGrad nodes bound to the current
+            # stream, and default-stream-bound AccumulateGrad breaks later
+            # graph captures (segfault in capture_end; torch warns about
+            # exactly this).
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                loss = self._eager_fwd_bwd(batch)
+            torch.cuda.current_stream().wait_stream(s)
             self._allreduce()
             self._clip_opt_ops()
             return loss
