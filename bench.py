@@ -87,25 +87,7 @@ def main() -> None:
 
     batch = {"x": x, "x_mask": x_mask, "y": y, "y_mask": y_mask}
     gstep = None
-    mgc = None
-    if on_gpu and args.graph != "off" and os.environ.get("WAP_USE_MGC"):
-        # torch-native graphed callables (captures fwd+bwd graphs with
-        # proper AccumulateGrad stream handling); opt/clip stay eager.
-        class _LossMod(torch.nn.Module):
-            def __init__(self, m):
-                super().__init__()
-                self.m = m
-
-            def forward(self, x, xm, y, ym):
-                with torch.autocast("cuda", dtype=torch.bfloat16,
-                                    enabled=use_bf16, cache_enabled=False):
-                    return self.m(x, xm, y, ym)["loss"]
-
-        for g in opt.param_groups:
-            g["capturable"] = True
-        mgc = torch.cuda.make_graphed_callables(
-            _LossMod(model), (x, x_mask, y, y_mask), num_warmup_iters=3)
-    if on_gpu and args.graph != "off" and mgc is None:
+    if on_gpu and args.graph != "off":
         try:
             from wap_amd.runtime.graph_train import GraphedTrainStep
             gstep = GraphedTrainStep(model, ddp, opt, clip=100.0,
@@ -120,14 +102,6 @@ def main() -> None:
             gstep = None
 
     def train_step():
-        if mgc is not None:
-            ddp.zero_grad()
-            loss = mgc(x, x_mask, y, y_mask)
-            loss.backward()
-            ddp.finish_gradient_sync()
-            torch.nn.utils.clip_grad_norm_(model.parameters(), 100.0)
-            opt.step()
-            return loss
         if gstep is not None:
             loss = gstep.step()
             if os.environ.get("WAP_GRAD_TRACE"):
@@ -228,7 +202,7 @@ def main() -> None:
                 "image": f"{args.height}x{args.width}",
                 "parallelism": f"dp{world}",
                 "kernels": ops.get_backend(),
-                "graph": gstep is not None or mgc is not None,
+                "graph": gstep is not None,
                 "channels_last": on_gpu and args.channels_last == "on",
             },
             "extra": extra,

@@ -48,11 +48,11 @@ class WAPConfig:
     device: str = "auto"               # auto | cpu | cuda
     dtype: str = "bf16"                # bf16 | fp32 (compute dtype on GPU)
     kernels: str = "auto"              # auto | eager | hip
-    # hipGraph-captured train steps: "auto" enables the multi-shape graphed
-    # trainer.  Default off: multiple backward-containing graph captures in
-    # one process segfault capture_end on ROCm 7.x/PyTorch 2.10 (the
-    # single-capture bench path is stable and stays graphed).
-    graph: str = "off"
+    # hipGraph-captured train steps (multi-shape trainer).  Requires the
+    # first backward of the process to run on a non-default stream (done by
+    # the trainer; default-stream AccumulateGrad binding breaks captures on
+    # ROCm 7.x) — see PERF.md for the full list of platform hazards.
+    graph: str = "auto"
     graph_pad_hw: int = 64             # bucket shape padding grid for graphs
     graph_pad_t: int = 16
     log_every: int = 20
