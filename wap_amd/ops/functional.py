@@ -103,7 +103,10 @@ class _MaskedSoftmaxContext(torch.autograd.Function):
             alpha, ctxv = _masked_softmax_context_eager(e_, a, mask_)
         ctx.save_for_backward(alpha, a)
         ctx.in_dtypes = (e.dtype, a.dtype)
-        return alpha, ctxv
+        # context in a's dtype: a f32 ctx would silently promote both
+        # decoder GRUs (and their gate GEMMs) to f32 under autocast —
+        # measured as thousands of f32 elementwise/cast kernels per step
+        return alpha, ctxv.to(a.dtype)
 
     @staticmethod
     def backward(ctx, g_alpha: Tensor, g_ctx: Tensor):
@@ -123,8 +126,9 @@ class _MaskedSoftmaxContext(torch.autograd.Function):
 
 
 def masked_softmax_context(e: Tensor, a: Tensor, mask: Tensor) -> Tuple[Tensor, Tensor]:
-    """(alpha[B,L] f32, ctx[B,D] f32) from scores e[B,L], annotations
-    a[B,L,D], mask[B,L] (1 = valid position)."""
+    """(alpha[B,L] f32, ctx[B,D] in a.dtype) from scores e[B,L],
+    annotations a[B,L,D], mask[B,L] (1 = valid position); softmax and the
+    context accumulation run in f32 internally."""
     return _MaskedSoftmaxContext.apply(e, a, mask)
 
 
