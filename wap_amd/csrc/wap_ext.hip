@@ -102,6 +102,8 @@ __global__ void zero_f32_kernel(float* __restrict__ p, long n) {
 // e[b,l] = sum_k nu[k] * tanh(q[b,k] + ua[b,l,k] + uf[b,l,k])
 // One wave per (b,l); q and nu staged in LDS per block.
 
+typedef __attribute__((ext_vector_type(8))) short bf16x8v;
+
 template <typename T>
 __global__ void attn_score_fwd_kernel(
     const T* __restrict__ q, const T* __restrict__ ua, const T* __restrict__ uf,
@@ -118,15 +120,39 @@ __global__ void attn_score_fwd_kernel(
   __syncthreads();
   const int lane = threadIdx.x & (WAVE - 1), w = threadIdx.x >> 6;
   const int nw = blockDim.x / WAVE;
+  // vector path: lane owns 8 CONSECUTIVE k per 512-chunk -> one 16-byte
+  // load per operand per chunk (vs 64 x 2-byte lane-strided loads)
+  const bool vec = std::is_same<T, __hip_bfloat16>::value && (A % 512 == 0);
   for (int l = blockIdx.y * nw + w; l < L; l += gridDim.y * nw) {
     const size_t off = ((size_t)b * L + l) * A;
     float acc = 0.f;
-    for (int k = lane; k < A; k += WAVE) {
-      const float t = tanhf(qs[k] + ldf(ua + off + k) + ldf(uf + off + k));
-      stf(tcache + off + k, t);
-      // accumulate the ROUNDED value (what the cache holds) so forward and
-      // backward see the same t, and so bf16 matches the eager oracle
-      acc += ns[k] * ldf(tcache + off + k);
+    if (vec) {
+      for (int base = 0; base < A; base += 512) {
+        const int k0 = base + lane * 8;
+        const bf16x8v va = *(const bf16x8v*)((const __hip_bfloat16*)ua + off + k0);
+        const bf16x8v vf = *(const bf16x8v*)((const __hip_bfloat16*)uf + off + k0);
+        bf16x8v tv;
+#pragma unroll
+        for (int ee = 0; ee < 8; ++ee) {
+          const short sa = va[ee], sf = vf[ee];
+          const __hip_bfloat16 au = *(const __hip_bfloat16*)&sa;
+          const __hip_bfloat16 fu = *(const __hip_bfloat16*)&sf;
+          const float t = tanhf(qs[k0 + ee] + __bfloat162float(au)
+                                + __bfloat162float(fu));
+          const __hip_bfloat16 tb = __float2bfloat16(t);
+          tv[ee] = *(const short*)&tb;
+          acc += ns[k0 + ee] * __bfloat162float(tb);
+        }
+        *(bf16x8v*)((__hip_bfloat16*)tcache + off + k0) = tv;
+      }
+    } else {
+      for (int k = lane; k < A; k += WAVE) {
+        const float t = tanhf(qs[k] + ldf(ua + off + k) + ldf(uf + off + k));
+        stf(tcache + off + k, t);
+        // accumulate the ROUNDED value (what the cache holds) so forward
+        // and backward see the same t as the eager oracle
+        acc += ns[k] * ldf(tcache + off + k);
+      }
     }
     acc = wave_sum(acc);
     if (lane == 0) e[(size_t)b * L + l] = acc;
