@@ -159,7 +159,12 @@ def main() -> None:
     extra = {"final_loss": final_loss}
     decode_rounds = args.decode_steps if args.decode_steps else (1 if on_gpu else 0)
     if rank == 0 and decode_rounds > 0 and not args.no_decode:
-        model.eval()
+        # Fixed-seed fresh model: hypothesis lengths (the tok/s numerator)
+        # must not depend on how far the timed training happened to get.
+        torch.manual_seed(4242)
+        dmodel = WAPModel(n_tokens=112, encoder=args.encoder).to(device)
+        dmodel.eval()
+        model = dmodel
         db = min(args.batch, 8)
         with autocast():
             beam_decode(model, x[:db], x_mask[:db], beam=args.beam, maxlen=32)
