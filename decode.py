@@ -21,7 +21,10 @@ import time
 import torch
 
 from wap_amd import ops
-from wap_amd.data import BucketLoader, Sample, Vocab, load_pairs, make_buckets, synth_samples
+from wap_amd.data import (
+    BucketLoader, Sample, Vocab, load_pairs, make_buckets, prepare_batch,
+    synth_samples,
+)
 from wap_amd.eval import score_pairs
 from wap_amd.models import WAPModel
 from wap_amd.runtime import beam_decode, load_checkpoint
@@ -70,7 +73,6 @@ def main() -> None:
     t0 = time.perf_counter()
     with open(args.out, "w", encoding="utf-8") as f:
         for bucket in buckets:
-            from wap_amd.data import prepare_batch
             batch = prepare_batch(bucket, device)
             with torch.autocast("cuda", dtype=torch.bfloat16, enabled=use_bf16):
                 hyps = beam_decode(model, batch["x"], batch["x_mask"],

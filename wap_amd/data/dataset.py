@@ -242,13 +242,16 @@ class BucketLoader:
         q: "queue.Queue" = queue.Queue(maxsize=2)
 
         def worker():
-            for i in order:
-                b = prepare_batch(self.local[i], "cpu", self.dtype,
-                                  pad_to=self.pad_to)
-                b = {k: (v.pin_memory() if torch.is_tensor(v) else v)
-                     for k, v in b.items()}
-                q.put(b)
-            q.put(None)
+            try:
+                for i in order:
+                    b = prepare_batch(self.local[i], "cpu", self.dtype,
+                                      pad_to=self.pad_to)
+                    b = {k: (v.pin_memory() if torch.is_tensor(v) else v)
+                         for k, v in b.items()}
+                    q.put(b)
+                q.put(None)
+            except BaseException as e:  # propagate instead of hanging main
+                q.put(e)
 
         t = threading.Thread(target=worker, daemon=True)
         t.start()
@@ -256,6 +259,8 @@ class BucketLoader:
             b = q.get()
             if b is None:
                 break
+            if isinstance(b, BaseException):
+                raise b
             yield {k: (v.to(dev, non_blocking=True) if torch.is_tensor(v) else v)
                    for k, v in b.items()}
         t.join()

@@ -6,7 +6,6 @@ Loss is the reference's masked token-level cross-entropy (SURVEY.md §3.1
 
 from __future__ import annotations
 
-from dataclasses import asdict
 from typing import Dict, Optional, Tuple
 
 import torch

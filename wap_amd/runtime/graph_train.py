@@ -18,7 +18,7 @@ between-graph all-reduce is one call per bucket, no packing.
 
 from __future__ import annotations
 
-from typing import Callable, Dict, Optional
+from typing import Dict, Optional
 
 import torch
 import torch.distributed as dist
