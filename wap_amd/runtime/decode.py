@@ -185,10 +185,23 @@ class BeamState:
 # does not invalidate a graph.  Bounded LRU.
 _GRAPH_CACHE: Dict[tuple, BeamState] = {}
 _GRAPH_CACHE_CAP = 16
+_MODEL_KEYS = 0
+
+
+def _model_key(model) -> int:
+    """Stable per-model cache id (id(model) can be recycled after GC while
+    cached graphs still hold the OLD model's weights)."""
+    global _MODEL_KEYS
+    k = getattr(model, "_wap_decode_key", None)
+    if k is None:
+        _MODEL_KEYS += 1
+        k = _MODEL_KEYS
+        model._wap_decode_key = k
+    return k
 
 
 def _graph_key(model, state: BeamState, fused: bool, autocast_on: bool):
-    return (id(model), state.b, state.k, state.hp, state.wp,
+    return (_model_key(model), state.b, state.k, state.hp, state.wp,
             state.S["a_flat"].shape[-1], str(state.S["a_flat"].dtype),
             fused, autocast_on)
 
