@@ -144,3 +144,19 @@ def test_pad_to_keeps_loss_exact():
         l1 = m(plain["x"], plain["x_mask"], plain["y"], plain["y_mask"])["loss"]
         l2 = m(padded["x"], padded["x_mask"], padded["y"], padded["y_mask"])["loss"]
     assert torch.allclose(l1, l2, atol=1e-5), (l1.item(), l2.item())
+
+
+def test_shard_more_ranks_than_buckets():
+    samples = _mk_samples(4)
+    from wap_amd.data import make_buckets, shard_buckets
+    buckets, _ = make_buckets(samples, batch_size=2, batch_imagesize=500000)
+    world = len(buckets) + 3
+    shards = [shard_buckets(buckets, r, world) for r in range(world)]
+    steps = {len(s) for s in shards}
+    assert steps == {0}  # all ranks agree on zero steps (no deadlock)
+
+
+def test_vocab_oov_raises(vocab):
+    import pytest as _pytest
+    with _pytest.raises(KeyError):
+        vocab.encode("definitely_not_a_token")

@@ -86,3 +86,28 @@ def test_beam_batch_matches_single():
     singles = [beam_decode(m, x[i:i+1], xm[i:i+1], beam=3, maxlen=12)[0]
                for i in range(3)]
     assert batched == singles
+
+
+def test_decode_all_masked_image_is_safe():
+    """A fully-padded image must decode without NaN/crash (alpha=0, ctx=0)."""
+    m = _toy_model().eval()
+    x = torch.zeros(2, 1, 32, 32)
+    xm = torch.ones(2, 32, 32)
+    xm[1] = 0  # image 1 entirely padding
+    hyps = beam_decode(m, x, xm, beam=3, maxlen=8)
+    assert len(hyps) == 2
+    assert all(len(h) <= 8 for h in hyps)
+
+
+def test_config_args_roundtrip():
+    import argparse
+    from wap_amd.config import WAPConfig, add_config_args, config_from_args
+    p = argparse.ArgumentParser()
+    add_config_args(p)
+    cfg = config_from_args(p.parse_args([
+        "--batch-size", "7", "--encoder", "vgg", "--graph", "off",
+        "--beam", "3",
+    ]))
+    assert cfg.batch_size == 7 and cfg.encoder == "vgg"
+    assert cfg.graph == "off" and cfg.beam == 3
+    assert cfg.maxlen == WAPConfig().maxlen  # defaults preserved

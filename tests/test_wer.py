@@ -44,3 +44,10 @@ def test_missing_hypothesis_counts_as_deletions(tmp_path):
     hyp.write_text("")
     r = score_files(str(ref), str(hyp))
     assert r.total_errors == 3 and r.exprate == 0
+
+
+def test_empty_reference_edge():
+    r = score_pairs([([], ["a", "b"])])
+    assert r.total_errors == 2 and r.exprate == 0.0
+    r2 = score_pairs([([], [])])
+    assert r2.total_errors == 0 and r2.exprate == 1.0
