@@ -111,3 +111,27 @@ def test_config_args_roundtrip():
     assert cfg.batch_size == 7 and cfg.encoder == "vgg"
     assert cfg.graph == "off" and cfg.beam == 3
     assert cfg.maxlen == WAPConfig().maxlen  # defaults preserved
+
+
+def test_beamstate_reset_from_reproduces_decode():
+    """reset_from must fully re-initialize a state: decoding problem B in
+    buffers previously used for problem A equals a fresh decode of B (the
+    decode graph cache relies on this)."""
+    from wap_amd.runtime.decode import BeamState
+    m = _toy_model().eval()
+    xa = torch.rand(2, 1, 32, 32)
+    xb = torch.rand(2, 1, 32, 32)
+    xm = torch.ones(2, 32, 32)
+    with torch.no_grad():
+        fresh = beam_decode(m, xb, xm, beam=3, maxlen=10, use_graph=False)
+        sa = BeamState(m, xa, xm, 3)
+        for _ in range(4):
+            sa.step()
+        sb = BeamState(m, xb, xm, 3)
+        sa.reset_from(sb)
+        for _ in range(10):
+            sa.step()
+            if sa.all_dead():
+                break
+        reused = sa.extract()
+    assert reused == fresh
