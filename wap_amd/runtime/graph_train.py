@@ -255,13 +255,15 @@ class MultiShapeGraphedTrainer:
         if self.g_opt is not None:
             self.g_opt.replay()
             return
-        # First step on every rank: capture the clip+opt graph.
+        # First use on every rank: capture the clip+opt graph, then replay
+        # it to actually apply THIS step's update (capture only records).
         safe_stream_sync()
         self.g_opt = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.g_opt, capture_error_mode=_capture_mode()):
             if self.clip > 0:
                 torch.nn.utils.clip_grad_norm_(self.model.parameters(), self.clip)
             self.opt.step()
+        self.g_opt.replay()
 
     def _eager_fwd_bwd(self, batch):
         for b in self.ddp.buckets:
