@@ -361,3 +361,127 @@ void bnrelu_conv_mfma_fp8_fwd_bk96(
   bnrelu_conv_mfma_fp8_body<96>(x, w, scale, shift, out_scale, out,
                                 Bn, H, W, C, KOUT, R, S, relu_on);
 }
+
+// ---------------------------------------------------------------------
+// Halo-tiled 3x3 conv (stride 1, pad 1), C == 96, KOUT <= 32.
+// The bottleneck 3x3 is always 96 -> 24; the current generic kernel
+// re-stages the input tile per tap (9 taps x 2 barriers).  Here one
+// workgroup owns TW consecutive x-positions of ONE row y and stages the
+// normalized 3-row halo [3][TW+2][96] ONCE (2 barriers total); all nine
+// taps read shifted LDS views.
+// Grid: (ceil(W/TW), B*H); 256 threads, 4 waves (wave w owns positions
+// [w*16, w*16+16) of the tile -> 1 m-frag x (KOUT/16) n-frags).
+// ---------------------------------------------------------------------
+
+#define TW3 64
+#define C3 96
+
+extern "C" __global__ __launch_bounds__(256, 2)
+void bnrelu_conv3_halo_fwd(
+    const bf16_t* __restrict__ x, const bf16_t* __restrict__ w,
+    const float* __restrict__ scale, const float* __restrict__ shift,
+    bf16_t* __restrict__ out,
+    int Bn, int H, int W, int KOUT, int relu_on) {
+  __shared__ bf16_t lA[3][TW3 + 2][C3];     // 38 KB
+  __shared__ bf16_t lB[9][32][C3];          // 54 KB (all taps, KOUT<=32)
+  __shared__ float lscale[C3], lshift[C3];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wv = tid >> 6;
+  const int x0 = blockIdx.x * TW3;
+  const int by = blockIdx.y;            // b*H + y
+  const int b = by / H, y = by % H;
+  const int NFRAG = (KOUT + 15) / 16;   // <= 2
+
+  if (tid < C3) {
+    lscale[tid] = scale[tid];
+    lshift[tid] = shift[tid];
+  }
+  __syncthreads();
+
+  // ---- stage the 3-row halo: 3*(TW3+2) = 198 positions x 96 ch.
+  // 198*12 = 2376 16B pieces; thread stages ceil(2376/256)=10 pieces.
+  {
+    const int PIECES = 3 * (TW3 + 2) * (C3 / 8);
+    for (int flat = tid; flat < PIECES; flat += blockDim.x) {
+      const int pc = flat % (C3 / 8);
+      const int pos = flat / (C3 / 8);
+      const int row = pos / (TW3 + 2);        // 0..2 -> y-1..y+1
+      const int xi = x0 + pos % (TW3 + 2) - 1;
+      const int yy = y + row - 1;
+      const int c0 = pc * 8;
+      float v[8];
+      if (yy >= 0 && yy < H && xi >= 0 && xi < W) {
+        const bf16x8 v8 = *(const bf16x8*)(x + (((long)b * H + yy) * W + xi) * C3 + c0);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const short sv = v8[e];
+          float t = bf2f(*(const bf16_t*)&sv) * lscale[c0 + e] + lshift[c0 + e];
+          if (relu_on) t = fmaxf(t, 0.f);
+          v[e] = t;
+        }
+      } else {
+#pragma unroll
+        for (int e = 0; e < 8; ++e) v[e] = 0.f;
+      }
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        lA[row][pos % (TW3 + 2)][c0 + e] = f2bf(v[e]);
+    }
+    // ---- stage ALL nine weight taps: 9*KOUT*96 <= 9*32*12 pieces
+    const int WP = 9 * 32 * (C3 / 8);
+    for (int flat = tid; flat < WP; flat += blockDim.x) {
+      const int pc = flat % (C3 / 8);
+      const int kk = (flat / (C3 / 8)) % 32;
+      const int tap = flat / (32 * (C3 / 8));
+      const int c0 = pc * 8;
+      if (kk < KOUT) {
+        *(bf16x8*)&lB[tap][kk][c0] =
+            *(const bf16x8*)(w + (((long)kk * 3 + tap / 3) * 3 + tap % 3) * C3 + c0);
+      } else {
+#pragma unroll
+        for (int e = 0; e < 8; ++e) lB[tap][kk][c0 + e] = f2bf(0.f);
+      }
+    }
+  }
+  __syncthreads();
+
+  f32x4 acc[1][2];
+#pragma unroll
+  for (int j = 0; j < 2; ++j) acc[0][j] = {0.f, 0.f, 0.f, 0.f};
+
+  // wave wv owns tile positions [wv*16, wv*16+16)
+#pragma unroll
+  for (int tap = 0; tap < 9; ++tap) {
+    const int dy = tap / 3, dx = tap % 3;
+#pragma unroll
+    for (int ks = 0; ks < C3 / 32; ++ks) {
+      const int kk = ks * 32 + (lane >> 4) * 8;
+      // A fragment: position = wv*16 + (lane&15); shifted x index dx
+      const bf16x8 afrag =
+          *(const bf16x8*)&lA[dy][wv * 16 + (lane & 15) + dx][kk];
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf) {
+        if (nf >= NFRAG) break;
+        const bf16x8 bfrag = *(const bf16x8*)&lB[tap][nf * 16 + (lane & 15)][kk];
+        acc[0][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag, bfrag, acc[0][nf], 0, 0, 0);
+      }
+    }
+  }
+
+#pragma unroll
+  for (int nf = 0; nf < 2; ++nf) {
+    if (nf >= NFRAG) break;
+    const int k = nf * 16 + (lane & 15);
+    if (k >= KOUT) continue;
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int xt = wv * 16 + (lane >> 4) * 4 + reg;   // tile position
+      const int xi = x0 + xt;
+      if (xi < W) out[(((long)b * H + y) * W + xi) * KOUT + k] =
+          f2bf(acc[0][nf][reg]);
+    }
+  }
+}

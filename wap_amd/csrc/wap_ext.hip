@@ -774,6 +774,9 @@ extern "C" __global__ void bnrelu_conv_mfma_fwd(
 extern "C" __global__ void bnrelu_conv_mfma_fwd_bk96(
     const __hip_bfloat16*, const __hip_bfloat16*, const float*, const float*,
     __hip_bfloat16*, int, int, int, int, int, int, int, int);
+extern "C" __global__ void bnrelu_conv3_halo_fwd(
+    const __hip_bfloat16*, const __hip_bfloat16*, const float*, const float*,
+    __hip_bfloat16*, int, int, int, int, int);
 extern "C" __global__ void bnrelu_conv_mfma_fp8_fwd(
     const __hip_fp8_e4m3*, const __hip_fp8_e4m3*, const float*, const float*,
     float, __hip_bfloat16*, int, int, int, int, int, int, int, int);
@@ -798,6 +801,19 @@ torch::Tensor bnrelu_conv_fwd(
               "bnrelu_conv: K<=96, odd kernel, same channels");
   auto out = at::empty({B, K, H, W}, x.options(),
                        at::MemoryFormat::ChannelsLast);
+  if (C == 96 && K <= 32 && R == 3 && S == 3) {
+    // halo-tiled 3x3: stage the normalized 3-row halo once, 2 barriers
+    // total (vs 2 per tap-chunk in the generic path)
+    const dim3 hgrid((W + 63) / 64, B * H);
+    hipLaunchKernelGGL(bnrelu_conv3_halo_fwd, hgrid, dim3(256), 0,
+                       cur_stream(),
+                       cptr<__hip_bfloat16>(x), cptr<__hip_bfloat16>(w),
+                       cptr<float>(scale), cptr<float>(shift),
+                       ptr<__hip_bfloat16>(out), B, H, W, K,
+                       relu_on ? 1 : 0);
+    HIP_OK();
+    return out;
+  }
   const long M = (long)B * H * W;
   const dim3 grid((M + 127) / 128);
   auto kern = (C % 96 == 0) ? bnrelu_conv_mfma_fwd_bk96 : bnrelu_conv_mfma_fwd;
