@@ -80,3 +80,36 @@ def test_world1_is_noop():
     m(x, xm, y, ym)["loss"].backward()
     ddp.finish_gradient_sync()  # must not raise without a process group
     assert all(p.grad is not None for p in m.parameters())
+
+
+def _frozen_worker(rank, world, port, out_dir):
+    from wap_amd import ops
+    ops.set_backend("eager")
+    dist.init_process_group(
+        "gloo", init_method=f"tcp://127.0.0.1:{port}", rank=rank, world_size=world
+    )
+    from wap_amd.parallel import BucketedDDP
+
+    m = _mk_model()
+    # freeze the embedding: its bucket's hook never fires; the
+    # finish_gradient_sync fallback must still reduce every bucket
+    m.parser.W_o.weight.requires_grad_(False)
+    ddp = BucketedDDP(m, bucket_bytes=1 << 16)
+    ddp.zero_grad()
+    x, xm, y, ym = _rank_batch(rank)
+    m(x, xm, y, ym)["loss"].backward()
+    ddp.finish_gradient_sync()  # must not deadlock or raise
+    torch.save({n: p.grad.clone() for n, p in m.named_parameters()
+                if p.requires_grad},
+               os.path.join(out_dir, f"fg_{rank}.pt"))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_frozen_param_buckets_still_sync(tmp_path):
+    port = 29733
+    mp.spawn(_frozen_worker, args=(2, port, str(tmp_path)), nprocs=2, join=True)
+    g0 = torch.load(tmp_path / "fg_0.pt", weights_only=False)
+    g1 = torch.load(tmp_path / "fg_1.pt", weights_only=False)
+    for n in g0:
+        assert torch.equal(g0[n], g1[n]), n

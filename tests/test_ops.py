@@ -127,3 +127,14 @@ def test_conv_1ch_matches_conv2d():
         assert torch.allclose(gx1, gx2, atol=1e-10)
         assert torch.allclose(gw1, gw2, atol=1e-10)
         assert torch.allclose(gb1, gb2, atol=1e-10)
+
+
+def test_ext_require_fails_loudly(monkeypatch):
+    """On a GPU box a missing extension must raise, never silently fall
+    back to eager (would fake benchmark results)."""
+    from wap_amd.ops import ext
+    monkeypatch.setattr(ext, "_tried", True)
+    monkeypatch.setattr(ext, "_mod", None)
+    monkeypatch.setattr(ext, "_err", ImportError("boom"))
+    with pytest.raises(RuntimeError, match="HIP extension is required"):
+        ext.require()
