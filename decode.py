@@ -60,7 +60,8 @@ def main() -> None:
         samples = [Sample(u, feats[u], vocab.encode(caps[u])) for u in feats]
     else:
         samples = load_pairs(args.test_pkl, args.test_captions, vocab)
-    buckets, _ = make_buckets(samples, args.batch_size, args.batch_imagesize)
+    buckets, dropped = make_buckets(samples, args.batch_size,
+                                    args.batch_imagesize)
 
     model = WAPModel(n_tokens=len(vocab), encoder=args.encoder).to(device)
     if args.ckpt:
@@ -87,6 +88,16 @@ def main() -> None:
                     [vocab.id_to_token[j] for j in ref_ids],
                     text.split(),
                 ))
+        # Over-size/over-length samples are not decodable under the image
+        # budget, but they must still appear in the output file and count as
+        # full deletions in the printed WER (so this score agrees with
+        # compute-wer.py run against the complete caption file).
+        for s in dropped:
+            f.write(f"{s.uid}\t\n")
+            pairs.append(([vocab.id_to_token[j] for j in s.label], []))
+    if dropped:
+        print(f"note: {len(dropped)} samples exceeded maxlen/max_imagesize; "
+              "emitted empty hypotheses for them")
     dt = time.perf_counter() - t0
     print(f"decoded {len(pairs)} expressions, {n_tok} tokens "
           f"in {dt:.2f}s ({n_tok / max(dt, 1e-9):.1f} tok/s) -> {args.out}")

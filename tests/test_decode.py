@@ -135,3 +135,51 @@ def test_beamstate_reset_from_reproduces_decode():
                 break
         reused = sa.extract()
     assert reused == fresh
+
+
+def test_banked_finished_hyp_survives_eviction():
+    """A hypothesis that emits <eos> and is later pushed out of the k-slot
+    beam must still be rankable at extract() time (death-event banking)."""
+    from wap_amd.runtime.decode import BeamState
+    st = BeamState.__new__(BeamState)
+    st.b, st.k = 1, 2
+    T = 3
+    # slot 0 dies at t=1 with a GOOD normalized score (-1.0/2 = -0.5); by the
+    # end both surviving slots are live continuations with worse norm scores.
+    st.tokens = [torch.tensor([[4, 5]]), torch.tensor([[EOS_ID, 6]]),
+                 torch.tensor([[7, 8]])]
+    st.parents = [torch.tensor([[0, 0]]), torch.tensor([[0, 1]]),
+                  torch.tensor([[1, 1]])]
+    st.dieds = [torch.zeros(1, 2, dtype=torch.bool) for _ in range(T)]
+    st.dieds[1][0, 0] = True
+    st.dscores = [torch.zeros(1, 2) for _ in range(T)]
+    st.dscores[1][0, 0] = -1.0
+    st.S = {
+        "scores": torch.tensor([[-9.0, -9.5]]),
+        "lengths": torch.tensor([[3, 3]]),
+        "alive": torch.tensor([[True, True]]),
+    }
+    out = st.extract()
+    # banked: backtrace from (t=1, slot 0): t1 slot0 tok=eos, parent 0;
+    # t0 slot0 tok=4 -> [4, eos] -> strip eos -> [4]
+    assert out == [[4]]
+
+
+def test_live_hyp_beats_worse_banked():
+    """Banking must not override a better live hypothesis."""
+    from wap_amd.runtime.decode import BeamState
+    st = BeamState.__new__(BeamState)
+    st.b, st.k = 1, 2
+    st.tokens = [torch.tensor([[4, 5]]), torch.tensor([[EOS_ID, 6]])]
+    st.parents = [torch.tensor([[0, 0]]), torch.tensor([[0, 1]])]
+    st.dieds = [torch.zeros(1, 2, dtype=torch.bool) for _ in range(2)]
+    st.dieds[1][0, 0] = True
+    st.dscores = [torch.zeros(1, 2) for _ in range(2)]
+    st.dscores[1][0, 0] = -8.0          # norm -4.0
+    st.S = {
+        "scores": torch.tensor([[-8.0, -1.0]]),   # slot 1 live, norm -0.5
+        "lengths": torch.tensor([[2, 2]]),
+        "alive": torch.tensor([[False, True]]),
+    }
+    out = st.extract()
+    assert out == [[5, 6]]

@@ -120,13 +120,15 @@ def main() -> None:
 
     log = MetricsLogger(cfg.metrics_jsonl, rank=rank)
     log.log("setup", world_size=world, device=str(device),
-            n_buckets=len(train_buckets), dropped=dropped,
+            n_buckets=len(train_buckets), dropped=len(dropped),
             encoder=cfg.encoder, kernels=ops.get_backend())
 
     model = WAPModel(
         n_tokens=len(vocab), encoder=cfg.encoder, embed_dim=cfg.embed_dim,
         hidden_dim=cfg.hidden_dim, attention_dim=cfg.attention_dim,
         coverage_ch=cfg.coverage_ch, coverage_kernel=cfg.coverage_kernel,
+        encoder_kw=({"growth": cfg.growth} if cfg.encoder == "densenet"
+                    else None),
     ).to(device)
     optimizer = build_optimizer(cfg, model)
 

@@ -611,8 +611,8 @@ std::vector<torch::Tensor> masked_softmax_context_bwd(
   auto de = torch::empty({B, L}, g_alpha.options().dtype(at::kFloat));
   auto da = torch::empty_like(a);
   auto sscr = torch::empty({B}, g_alpha.options().dtype(at::kFloat));
-  hipLaunchKernelGGL(zero_f32_kernel, dim3(1), dim3(256), 0, cur_stream(),
-                     ptr<float>(sscr), (long)B);
+  hipLaunchKernelGGL(zero_f32_kernel, dim3((B + 255) / 256), dim3(256), 0,
+                     cur_stream(), ptr<float>(sscr), (long)B);
   const int smem = (16 + D) * sizeof(float);
   const dim3 grid((L + MSC_BWD_LCH - 1) / MSC_BWD_LCH, B);
   if (is_bf16(a)) {

@@ -84,16 +84,19 @@ def make_buckets(
     batch_imagesize: int = 500000,
     maxlen: int = 200,
     max_imagesize: int = 500000,
-) -> Tuple[List[Bucket], int]:
+) -> Tuple[List[Bucket], List[Sample]]:
     """Sort by image area, pack greedily under the pixel budget and the batch
     size cap; drop over-size/over-length samples.  Returns (buckets,
-    n_dropped).  Deterministic for a given input order."""
+    dropped_samples) so callers can count AND report the dropped uids (a
+    decode entrypoint must still emit a line per dropped uid — otherwise its
+    printed WER and compute-wer.py on the full caption file disagree).
+    Deterministic for a given input order."""
     kept = []
-    dropped = 0
+    dropped: List[Sample] = []
     for s in samples:
         h, w = s.image.shape
         if len(s.label) > maxlen or h * w > max_imagesize:
-            dropped += 1
+            dropped.append(s)
             continue
         kept.append(s)
     kept.sort(key=lambda s: (s.image.shape[0] * s.image.shape[1], s.uid))
@@ -117,25 +120,36 @@ def make_buckets(
 
 
 def shard_buckets(buckets: List[Bucket], rank: int, world_size: int) -> List[Bucket]:
-    """Deal whole buckets to ranks so per-rank pixel totals stay balanced:
-    sort by padded pixel cost descending, then greedy-assign each bucket to
-    the currently lightest rank (deterministic)."""
+    """Deal whole buckets to ranks so per-rank pixel totals stay balanced AND
+    every rank runs the same number of steps per epoch (collectives!).
+
+    Greedy: sort by padded pixel cost descending, assign each bucket to the
+    lightest rank that still has room under ceil(n/world) buckets.  Ranks
+    left short by the remainder are padded with duplicates of the smallest
+    buckets (DistributedSampler-style) — no data is ever dropped, a few
+    samples are seen twice per epoch instead.  Deterministic."""
     if world_size <= 1:
         return list(buckets)
+    if not buckets:
+        return []
     order = sorted(range(len(buckets)),
                    key=lambda i: (-buckets[i].pixels, i))
+    cap = (len(buckets) + world_size - 1) // world_size
     loads = [0] * world_size
-    mine: List[Bucket] = []
-    counts = [0] * world_size
+    assign: List[List[int]] = [[] for _ in range(world_size)]
     for i in order:
-        r = min(range(world_size), key=lambda j: (loads[j], counts[j], j))
+        r = min((j for j in range(world_size) if len(assign[j]) < cap),
+                key=lambda j: (loads[j], len(assign[j]), j))
         loads[r] += buckets[i].pixels
-        counts[r] += 1
-        if r == rank:
-            mine.append(buckets[i])
-    # All ranks must run the same number of steps per epoch (collectives!).
-    n_steps = min(counts) if min(counts) > 0 else 0
-    return mine[:n_steps]
+        assign[r].append(i)
+    # Pad short ranks with the globally smallest buckets (duplicates).
+    pad_src = list(reversed(order))  # smallest first
+    pi = 0
+    for r in range(world_size):
+        while len(assign[r]) < cap:
+            assign[r].append(pad_src[pi % len(pad_src)])
+            pi += 1
+    return [buckets[i] for i in assign[rank]]
 
 
 def _round_up(v: int, m: int) -> int:

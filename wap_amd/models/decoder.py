@@ -127,7 +127,9 @@ class Parser(nn.Module):
         a_flat, ua = self.attention.precompute(a)
         mask_flat = mask.flatten(1)
         s = self.init_state(a_flat, mask_flat)
-        coverage = torch.zeros(b, hp, wp, device=a.device, dtype=a.dtype)
+        # f32 from step 0 (alpha is f32, so step>=1 coverage is f32 anyway);
+        # decode's BeamState uses the same dtype so the two paths match
+        coverage = torch.zeros(b, hp, wp, device=a.device, dtype=torch.float32)
         emb_prev = torch.zeros(b, self.embedding.embedding_dim,
                                device=a.device, dtype=a.dtype)
         logits = []

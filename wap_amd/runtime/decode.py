@@ -70,6 +70,10 @@ def _beam_step_(model, S: Dict[str, Tensor], b: int, k: int, fused: bool) -> Non
     S["coverage"].copy_(new_cov[rows])
     parent_alive = torch.gather(S["alive"], 1, parent)
     S["lengths"].copy_(torch.gather(S["lengths"], 1, parent) + parent_alive.long())
+    # Bank finished hypotheses: a hyp that emits <eos> here could later be
+    # evicted from the k-slot beam by a temporarily-better live candidate;
+    # recording the death event lets extract() still rank it.
+    S["died"].copy_(parent_alive & (token == EOS_ID))
     S["alive"].copy_(parent_alive & (token != EOS_ID))
     S["scores"].copy_(top_scores)
     S["token"].copy_(token)
@@ -96,11 +100,15 @@ class BeamState:
             "ua": _expand_for_beams(ua, beam).contiguous(),
             "mask": mask_bk.contiguous(),
             "s": s0.contiguous(),
-            "coverage": torch.zeros(b * beam, hp, wp, device=dev, dtype=a.dtype),
+            # coverage accumulates in f32, matching the training path (the
+            # teacher-forced unroll promotes coverage to f32 at step 1)
+            "coverage": torch.zeros(b * beam, hp, wp, device=dev,
+                                    dtype=torch.float32),
             "emb": torch.zeros(b * beam, model.parser.embedding.embedding_dim,
                                device=dev, dtype=s0.dtype),
             "scores": torch.full((b, beam), NEG_INF, device=dev),
             "alive": torch.ones(b, beam, dtype=torch.bool, device=dev),
+            "died": torch.zeros(b, beam, dtype=torch.bool, device=dev),
             "lengths": torch.zeros(b, beam, dtype=torch.long, device=dev),
             "token": torch.zeros(b, beam, dtype=torch.long, device=dev),
             "parent": torch.zeros(b, beam, dtype=torch.long, device=dev),
@@ -109,6 +117,8 @@ class BeamState:
         self.S["scores"][:, 0] = 0.0
         self.tokens: List[Tensor] = []     # [B, k] per step
         self.parents: List[Tensor] = []    # [B, k] back-pointers
+        self.dieds: List[Tensor] = []      # [B, k] finished-here flags
+        self.dscores: List[Tensor] = []    # [B, k] scores at that step
         self.graph = None
 
     def reset_from(self, other: "BeamState") -> None:
@@ -121,8 +131,10 @@ class BeamState:
         S["scores"].fill_(NEG_INF)
         S["scores"][:, 0] = 0.0
         S["alive"].fill_(True)
+        S["died"].fill_(False)
         S["lengths"].zero_()
         self.tokens, self.parents = [], []
+        self.dieds, self.dscores = [], []
 
     def step(self, fused: bool = True) -> None:
         _beam_step_(self.model, self.S, self.b, self.k, fused)
@@ -131,6 +143,8 @@ class BeamState:
     def record(self) -> None:
         self.tokens.append(self.S["token"].clone())
         self.parents.append(self.S["parent"].clone())
+        self.dieds.append(self.S["died"].clone())
+        self.dscores.append(self.S["scores"].clone())
 
     # ---- hipGraph capture -------------------------------------------------
     def capture(self, fused: bool = True, autocast_bf16: bool = False) -> None:
@@ -159,17 +173,41 @@ class BeamState:
         return bool((~self.S["alive"]).all().item())
 
     def extract(self) -> List[List[int]]:
-        """Backtrace the best (length-normalized) hypothesis per image."""
-        b, k = self.b, self.k
-        norm = self.S["scores"] / self.S["lengths"].clamp_min(1).float()
-        best = norm.argmax(dim=1).cpu()
-        out: List[List[int]] = []
+        """Best (length-normalized) hypothesis per image.
+
+        Candidates are the union of (a) banked death events (score frozen at
+        the step the hyp emitted <eos>, length = that step + 1) — so a
+        finished hyp evicted from the beam later is still considered — and
+        (b) the live hypotheses surviving at the last step; ranked together
+        by length-normalized score.  (Surviving dead slots are a subset of
+        their own death events.)"""
+        b = self.b
         tokens = [t.cpu() for t in self.tokens]
         parents = [p.cpu() for p in self.parents]
+        dieds = [d.cpu() for d in self.dieds]
+        dscores = [s.cpu() for s in self.dscores]
+        live_norm = (self.S["scores"]
+                     / self.S["lengths"].clamp_min(1).float()).cpu()
+        alive = self.S["alive"].cpu()
+        T = len(tokens)
+        out: List[List[int]] = []
         for i in range(b):
-            beam_idx = int(best[i])
+            best = None  # (norm_score, step, slot)
+            for t in range(T):
+                d = dieds[t][i]
+                if d.any():
+                    for slot in d.nonzero().flatten().tolist():
+                        sc = float(dscores[t][i, slot]) / (t + 1)
+                        if best is None or sc > best[0]:
+                            best = (sc, t, slot)
+            for slot in range(self.k):
+                if alive[i, slot] or best is None:
+                    sc = float(live_norm[i, slot])
+                    if best is None or sc > best[0]:
+                        best = (sc, T - 1, slot)
+            _, t0, beam_idx = best
             seq: List[int] = []
-            for t in range(len(tokens) - 1, -1, -1):
+            for t in range(t0, -1, -1):
                 seq.append(int(tokens[t][i, beam_idx]))
                 beam_idx = int(parents[t][i, beam_idx])
             seq.reverse()
