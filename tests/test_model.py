@@ -90,3 +90,37 @@ def test_attention_normalized_and_masked():
     assert torch.allclose(alpha.sum(1), torch.ones(2), atol=1e-5)
     am = alpha.view(2, a.shape[2], a.shape[3])
     assert (am[0, :, 2:] == 0).all()  # no attention on padding
+
+
+def test_forward_matches_stepwise_rollout():
+    """The batched teacher-forced unroll (forward) must produce the same
+    logits as composing parser.step() per timestep — the decode path and the
+    training path share per-step math."""
+    import torch
+    from wap_amd.models import WAPModel
+    torch.manual_seed(11)
+    m = WAPModel(n_tokens=13, encoder="vgg", embed_dim=12, hidden_dim=12,
+                 attention_dim=16, coverage_ch=8, coverage_kernel=3).eval()
+    b, T = 3, 7
+    x = torch.rand(b, 1, 32, 48)
+    xm = torch.ones(b, 32, 48)
+    xm[1, :, 30:] = 0
+    y = torch.randint(1, 13, (b, T))
+    with torch.no_grad():
+        a, mask = m.encode(x, xm)
+        batched = m.parser(a, mask, y)
+        # manual per-step rollout
+        a_flat, ua = m.parser.attention.precompute(a)
+        s = m.parser.init_state(a_flat, mask.flatten(1))
+        hp, wp = a.shape[2], a.shape[3]
+        cov = torch.zeros(b, hp, wp)
+        emb = torch.zeros(b, 12)
+        outs = []
+        for t in range(T):
+            logits, s, _, alpha = m.parser.step(emb, s, a_flat, ua, cov, mask)
+            outs.append(logits)
+            cov = cov + alpha.view(b, hp, wp)
+            emb = m.parser.embedding(y[:, t])
+        stepwise = torch.stack(outs, dim=1)
+    assert torch.allclose(batched, stepwise, atol=1e-5), \
+        (batched - stepwise).abs().max().item()

@@ -15,7 +15,7 @@ run through ``wap_amd.ops`` (HIP-fused on gfx950).
 
 from __future__ import annotations
 
-from typing import Tuple
+from typing import Optional, Tuple
 
 import torch
 from torch import Tensor, nn
@@ -54,6 +54,31 @@ class CoverageAttention(nn.Module):
         w = self.U_f.weight @ self.conv_Q.weight.view(q_ch, kh * kw)
         return w.view(self.attention_dim, 1, kh, kw)
 
+    def attend(
+        self,
+        q: Tensor,          # [B, A] precomputed W_a s_hat
+        a_flat: Tensor,     # [B, L, D]
+        ua: Tensor,         # [B, L, A] precomputed U_a a
+        coverage: Tensor,   # [B, H', W'] running sum of alphas
+        mask: Tensor,       # [B, H', W'] feature-grid mask
+        w_cov: Optional[Tensor] = None,  # merged_coverage_weight(), hoisted
+    ) -> Tuple[Tensor, Tensor]:
+        """Attention with the query GEMM and coverage-weight merge done by
+        the caller (the training unroll batches/hoists them across steps).
+        Returns (context[B,D], alpha[B,L])."""
+        b, hp, wp = coverage.shape
+        w = self.merged_coverage_weight() if w_cov is None else w_cov
+        kh = w.shape[2]
+        # 1-channel conv via unfold+GEMM (MIOpen naive-kernel bypass, see
+        # encoder.conv_1ch) — this runs once per decode timestep.
+        from .encoder import conv_1ch
+        uf = conv_1ch(coverage.unsqueeze(1), w, padding=kh // 2)  # [B,A,H',W']
+        uf = uf.flatten(2).transpose(1, 2) + self.U_f.bias        # [B,L,A]
+        alpha, ctx = ops.coverage_attention(
+            q, ua, uf, self.nu, a_flat, mask.flatten(1)
+        )
+        return ctx, alpha
+
     def forward(
         self,
         s_hat: Tensor,      # [B, n] pre-attention decoder state
@@ -61,18 +86,8 @@ class CoverageAttention(nn.Module):
         ua: Tensor,         # [B, L, A] precomputed U_a a
         coverage: Tensor,   # [B, H', W'] running sum of alphas
         mask: Tensor,       # [B, H', W'] feature-grid mask
+        w_cov: Optional[Tensor] = None,
     ) -> Tuple[Tensor, Tensor]:
         """Returns (context[B,D], alpha[B,L])."""
-        b, hp, wp = coverage.shape
-        w = self.merged_coverage_weight()
-        kh = w.shape[2]
-        # 1-channel conv via unfold+GEMM (MIOpen naive-kernel bypass, see
-        # encoder.conv_1ch) — this runs once per decode timestep.
-        from .encoder import conv_1ch
-        uf = conv_1ch(coverage.unsqueeze(1), w, padding=kh // 2)  # [B,A,H',W']
-        uf = uf.flatten(2).transpose(1, 2) + self.U_f.bias        # [B,L,A]
         q = self.W_a(s_hat)                                 # [B,A]
-        alpha, ctx = ops.coverage_attention(
-            q, ua, uf, self.nu, a_flat, mask.flatten(1)
-        )
-        return ctx, alpha
+        return self.attend(q, a_flat, ua, coverage, mask, w_cov)

@@ -85,17 +85,19 @@ class Parser(nn.Module):
         return torch.tanh(self.W_init(mean_a))
 
     def readout(self, s: Tensor, c: Tensor, ey: Tensor) -> Tensor:
-        """Maxout readout -> logits[B,K].
+        """Maxout readout -> logits[..., K] (any leading dims).
 
         W_s s + W_c c + W_y ey == [s|c|ey] @ [W_s|W_c|W_y]^T: one GEMM
-        instead of three + two adds (the decode loop is launch-bound)."""
+        instead of three + two adds.  The training unroll calls this ONCE
+        over the stacked [B,T,...] states (the readout does not feed back
+        into the recurrence), so its forward+backward GEMMs and the weight
+        concatenations happen once per sequence, not once per step."""
         w_cat = torch.cat([self.W_s.weight, self.W_c.weight, self.W_y.weight],
                           dim=1)
-        x_cat = torch.cat([s, c.to(s.dtype), ey.to(s.dtype)], dim=1)
+        x_cat = torch.cat([s, c.to(s.dtype), ey.to(s.dtype)], dim=-1)
         pre = torch.nn.functional.linear(x_cat, w_cat, self.W_s.bias)
-        b = pre.shape[0]
-        pre = pre.view(b, self.readout_dim, self.maxout_pieces)
-        return self.W_o(pre.max(dim=2).values)
+        pre = pre.view(*pre.shape[:-1], self.readout_dim, self.maxout_pieces)
+        return self.W_o(pre.max(dim=-1).values)
 
     def step(
         self,
@@ -106,12 +108,13 @@ class Parser(nn.Module):
         coverage: Tensor,     # [B, H', W']
         mask: Tensor,         # [B, H', W']
         fused: bool = False,
+        w_cov: Optional[Tensor] = None,
     ) -> Tuple[Tensor, Tensor, Tensor, Tensor]:
         """One decode step -> (logits, s_t, c_t, alpha)."""
         gru1 = self.gru1.step_fused if fused else self.gru1
         gru2 = self.gru2.step_fused if fused else self.gru2
         s_hat = gru1(y_prev_emb, s_prev)
-        c_t, alpha = self.attention(s_hat, a_flat, ua, coverage, mask)
+        c_t, alpha = self.attention(s_hat, a_flat, ua, coverage, mask, w_cov)
         s_t = gru2(c_t, s_hat)
         logits = self.readout(s_t, c_t, y_prev_emb)
         return logits, s_t, c_t, alpha
@@ -122,7 +125,21 @@ class Parser(nn.Module):
         mask: Tensor,       # [B, H', W']
         y: Tensor,          # [B, T] target tokens (teacher forcing)
     ) -> Tensor:
-        """Teacher-forced unroll -> logits[B, T, K]."""
+        """Teacher-forced unroll -> logits[B, T, K].
+
+        Launch-count-aware restructuring of the per-step math (the unroll is
+        the dispatch-bound part of the train step; see PERF.md):
+
+        - the GRU1 input gates depend only on the (known) teacher-forced
+          tokens: ONE [B*T] GEMM up front instead of one per step;
+        - the attention query W_a s_hat and GRU2's h-side gates both consume
+          s_hat: merged into one GEMM per step;
+        - the merged 11x11 coverage-conv weight is built once per sequence;
+        - the maxout readout does not feed the recurrence: it runs ONCE over
+          the stacked states after the loop (one fwd GEMM + one bwd GEMM
+          pair instead of T of each).
+
+        Per-step math is identical to :meth:`step` (tested)."""
         b, _, hp, wp = a.shape
         a_flat, ua = self.attention.precompute(a)
         mask_flat = mask.flatten(1)
@@ -130,15 +147,29 @@ class Parser(nn.Module):
         # f32 from step 0 (alpha is f32, so step>=1 coverage is f32 anyway);
         # decode's BeamState uses the same dtype so the two paths match
         coverage = torch.zeros(b, hp, wp, device=a.device, dtype=torch.float32)
-        emb_prev = torch.zeros(b, self.embedding.embedding_dim,
-                               device=a.device, dtype=a.dtype)
-        logits = []
         T = y.shape[1]
+        att = self.attention
+        # teacher-forced embeddings for steps 0..T-1 (step 0 = zero "token")
+        emb = self.embedding(y[:, : T - 1])
+        emb_all = torch.cat([emb.new_zeros(b, 1, emb.shape[-1]), emb], dim=1)
+        gx1_all = torch.nn.functional.linear(emb_all, self.gru1.w_x,
+                                             self.gru1.b_x)       # [B,T,3n]
+        A = att.attention_dim
+        w_qh = torch.cat([att.W_a.weight, self.gru2.w_h], dim=0)  # [A+3n, n]
+        b_qh = torch.cat([self.gru2.b_h.new_zeros(A), self.gru2.b_h])
+        w_cov = att.merged_coverage_weight()
+        ss, cs = [], []
         for t in range(T):
-            step_logits, s, _, alpha = self.step(
-                emb_prev, s, a_flat, ua, coverage, mask
-            )
-            logits.append(step_logits)
+            gh1 = torch.nn.functional.linear(s, self.gru1.w_h, self.gru1.b_h)
+            s_hat = ops.gru_gates(gx1_all[:, t], gh1, s)
+            qh = torch.nn.functional.linear(s_hat, w_qh, b_qh)
+            c_t, alpha = att.attend(qh[:, :A], a_flat, ua, coverage, mask,
+                                    w_cov)
+            gx2 = torch.nn.functional.linear(c_t, self.gru2.w_x, self.gru2.b_x)
+            s = ops.gru_gates(gx2, qh[:, A:], s_hat)
             coverage = coverage + alpha.view(b, hp, wp)
-            emb_prev = self.embedding(y[:, t])
-        return torch.stack(logits, dim=1)
+            ss.append(s)
+            cs.append(c_t)
+        s_all = torch.stack(ss, dim=1)                            # [B,T,n]
+        c_all = torch.stack(cs, dim=1)                            # [B,T,D]
+        return self.readout(s_all, c_all, emb_all)

@@ -65,6 +65,7 @@ from .functional import (  # noqa: E402
     coverage_attention,
     fused_gru_cell,
     gru_cell,
+    gru_gates,
     masked_softmax_context,
     unfold_1ch,
 )
@@ -72,5 +73,5 @@ from .functional import (  # noqa: E402
 __all__ = [
     "set_backend", "get_backend", "use_hip",
     "attention_score", "coverage_attention", "masked_softmax_context",
-    "gru_cell", "fused_gru_cell", "unfold_1ch",
+    "gru_cell", "gru_gates", "fused_gru_cell", "unfold_1ch",
 ]

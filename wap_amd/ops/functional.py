@@ -200,6 +200,15 @@ def gru_cell(
     return _GruGates.apply(gx, gh, h)
 
 
+def gru_gates(gx: Tensor, gh: Tensor, h: Tensor) -> Tensor:
+    """GRU step from PRECOMPUTED gate projections (gates packed [r|z|n]).
+
+    The training unroll batches the x-side GEMM over all T teacher-forced
+    steps and merges the h-side GEMM with the attention query GEMM, so only
+    the fused nonlinearity remains per step."""
+    return _GruGates.apply(gx, gh, h)
+
+
 class _GruGates(torch.autograd.Function):
     """Fused gate nonlinearity + blend (everything after the two GEMMs)."""
 

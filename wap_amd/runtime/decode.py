@@ -50,7 +50,7 @@ def _beam_step_(model, S: Dict[str, Tensor], b: int, k: int, fused: bool) -> Non
     hp, wp = S["coverage"].shape[1], S["coverage"].shape[2]
     logits, s_new, _, alpha = m.step(
         S["emb"], S["s"], S["a_flat"], S["ua"], S["coverage"], S["mask"],
-        fused=fused,
+        fused=fused, w_cov=S["w_cov"],
     )
     logprob = torch.log_softmax(logits.float(), dim=-1).view(b, k, -1)
     K = logprob.shape[-1]
@@ -96,6 +96,11 @@ class BeamState:
         a_flat_bk = _expand_for_beams(a_flat, beam).contiguous()
         s0 = model.parser.init_state(a_flat_bk, mask_bk.flatten(1))
         self.S: Dict[str, Tensor] = {
+            # merged coverage-conv weight, built once per decode problem
+            # (derived from live weights; reset_from refreshes it so cached
+            # graphs never replay a stale copy after training steps)
+            "w_cov": model.parser.attention.merged_coverage_weight()
+                     .detach().contiguous(),
             "a_flat": a_flat_bk,
             "ua": _expand_for_beams(ua, beam).contiguous(),
             "mask": mask_bk.contiguous(),
@@ -126,7 +131,7 @@ class BeamState:
         the same shape: copy the per-problem tensors in, reset the mutable
         state.  The hipGraph stays valid (same storage)."""
         S, O = self.S, other.S
-        for n in ("a_flat", "ua", "mask", "s", "coverage", "emb"):
+        for n in ("a_flat", "ua", "mask", "s", "coverage", "emb", "w_cov"):
             S[n].copy_(O[n])
         S["scores"].fill_(NEG_INF)
         S["scores"][:, 0] = 0.0
