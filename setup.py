@@ -21,7 +21,8 @@ setup(
     ext_modules=[
         CUDAExtension(
             name="wap_amd._wap_hip",
-            sources=["wap_amd/csrc/wap_ext.hip", "wap_amd/csrc/conv_mfma.hip"],
+            sources=["wap_amd/csrc/wap_ext.hip", "wap_amd/csrc/conv_mfma.hip",
+                     "wap_amd/csrc/fattn.hip"],
             extra_compile_args={
                 "cxx": ["-O3"],
                 "nvcc": ["-O3", "-std=c++17"],

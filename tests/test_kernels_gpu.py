@@ -270,3 +270,90 @@ def test_bnrelu_conv_fp8_vs_fp32(C, K, R, HW):
     assert out.shape == ref.shape
     # remaining error: activation re-quantization to e4m3 inside the kernel
     assert _rel_err(out, ref) < 6e-2, _rel_err(out, ref)
+
+
+@pytest.mark.parametrize("B,Hp,Wp,A,KH", [
+    (3, 8, 32, 512, 11),   # headline train/decode shape
+    (2, 7, 19, 512, 11),   # odd grid
+    (2, 6, 10, 448, 5),    # non-default A, small kernel
+])
+def test_fused_cov_score_fwd_bwd(B, Hp, Wp, A, KH):
+    """Single-launch fused coverage conv + score (fattn.hip) vs the fp32
+    eager composition (conv via unfold+matmul, then score), fwd and bwd."""
+    from wap_amd.ops import functional as opf
+    torch.manual_seed(5)
+    dev = "cuda"
+    L = Hp * Wp
+    KK = KH * KH
+
+    def make(requires=True):
+        torch.manual_seed(7)
+        cov = torch.rand(B, Hp, Wp, device=dev) * 2
+        w = torch.randn(A, 1, KH, KH, device=dev) * 0.1
+        bias = torch.randn(A, device=dev) * 0.1
+        q = torch.randn(B, A, device=dev)
+        ua = torch.randn(B, L, A, device=dev)
+        nu = torch.randn(A, device=dev) * 0.3
+        for t in (cov, w, bias, q, ua, nu):
+            t.requires_grad_(requires)
+        return cov, w, bias, q, ua, nu
+
+    # fp32 eager oracle (same math, unfused)
+    cov, w, bias, q, ua, nu = make()
+    uf = torch.nn.functional.conv2d(cov.unsqueeze(1), w, padding=KH // 2)
+    uf = uf.flatten(2).transpose(1, 2) + bias
+    t = torch.tanh(q.unsqueeze(1) + ua + uf)
+    e_ref = t @ nu
+    torch.manual_seed(9)
+    gout = torch.randn_like(e_ref)
+    e_ref.backward(gout)
+    refs = (e_ref, cov.grad, w.grad, bias.grad, q.grad, ua.grad, nu.grad)
+
+    # fused bf16 kernel
+    cov2, w2, bias2, q2, ua2, nu2 = make()
+    covh = cov2.detach().requires_grad_()
+    qh = q2.detach().to(torch.bfloat16).requires_grad_()
+    uah = ua2.detach().to(torch.bfloat16).requires_grad_()
+    assert opf.can_fuse_cov_score(covh, w2, uah)
+    e_hip = opf._FusedCovScore.apply(
+        covh, w2, *opf.pack_coverage_weight(w2), bias2, qh, uah, nu2)
+    torch.manual_seed(9)
+    e_hip.backward(torch.randn_like(e_hip))
+    hips = (e_hip, covh.grad, w2.grad, bias2.grad, qh.grad, uah.grad,
+            nu2.grad)
+
+    names = ("e", "dcov", "dw", "dbias", "dq", "dua", "dnu")
+    budgets = (4e-2, 6e-2, 6e-2, 5e-2, 5e-2, 5e-2, 5e-2)
+    for name, h, r, budget in zip(names, hips, refs, budgets):
+        err = _rel_err(h, r)
+        assert err < budget, (name, err)
+
+
+def test_fused_cov_score_matches_unfused_hip():
+    """The fused path and the kernel-composed unfused path must agree at
+    bf16 tolerance through the full attend() (alpha and context)."""
+    import os
+    from wap_amd.models.attention import CoverageAttention
+    torch.manual_seed(13)
+    att = CoverageAttention(96, 32, 512, 512, 11).cuda()
+    B, Hp, Wp = 2, 8, 24
+    a = torch.randn(B, 96, Hp, Wp, device="cuda")
+    mask = torch.ones(B, Hp, Wp, device="cuda")
+    mask[1, :, 17:] = 0
+    cov = torch.rand(B, Hp, Wp, device="cuda")
+    s_hat = torch.randn(B, 32, device="cuda")
+    with torch.autocast("cuda", dtype=torch.bfloat16):
+        a_flat, ua = att.precompute(a)
+        ctx1, al1 = att(s_hat, a_flat, ua, cov, mask)
+    os.environ["WAP_EAGER_OPS"] = ""
+    from wap_amd import ops as _ops
+    from wap_amd.ops import functional as opf
+    old = opf.op_eager
+    try:
+        opf.op_eager = lambda n: n == "fattn"
+        with torch.autocast("cuda", dtype=torch.bfloat16):
+            ctx2, al2 = att(s_hat, a_flat, ua, cov, mask)
+    finally:
+        opf.op_eager = old
+    assert _rel_err(ctx1, ctx2) < 3e-2
+    assert _rel_err(al1, al2) < 5e-2

@@ -856,7 +856,17 @@ torch::Tensor bnrelu_conv_fwd_fp8(
   return out;
 }
 
+// fused coverage-attention score (fattn.hip)
+std::vector<torch::Tensor> fattn_score_fwd(
+    torch::Tensor cov, torch::Tensor wpad, torch::Tensor q, torch::Tensor ua,
+    torch::Tensor nu, torch::Tensor bias, int64_t KH, int64_t KW);
+std::vector<torch::Tensor> fattn_score_bwd(
+    torch::Tensor de, torch::Tensor tcache, torch::Tensor nu,
+    torch::Tensor wTpad, int64_t Hp, int64_t Wp, int64_t KH, int64_t KW);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("fattn_score_fwd", &fattn_score_fwd);
+  m.def("fattn_score_bwd", &fattn_score_bwd);
   m.def("bnrelu_conv_fwd", &bnrelu_conv_fwd);
   m.def("bnrelu_conv_fwd_fp8", &bnrelu_conv_fwd_fp8);
   m.def("attention_score_fwd", &attention_score_fwd);

@@ -62,6 +62,7 @@ class CoverageAttention(nn.Module):
         coverage: Tensor,   # [B, H', W'] running sum of alphas
         mask: Tensor,       # [B, H', W'] feature-grid mask
         w_cov: Optional[Tensor] = None,  # merged_coverage_weight(), hoisted
+        w_pack=None,        # ops.pack_coverage_weight(w_cov), hoisted
     ) -> Tuple[Tensor, Tensor]:
         """Attention with the query GEMM and coverage-weight merge done by
         the caller (the training unroll batches/hoists them across steps).
@@ -69,6 +70,15 @@ class CoverageAttention(nn.Module):
         b, hp, wp = coverage.shape
         w = self.merged_coverage_weight() if w_cov is None else w_cov
         kh = w.shape[2]
+        from ..ops import functional as opf
+        if opf.can_fuse_cov_score(coverage, w, ua):
+            # single-launch fused path (csrc/fattn.hip): conv-over-coverage
+            # + bias + additive score in one kernel, fwd and bwd
+            alpha, ctx = opf.fused_cov_attention(
+                q, ua, coverage, w, w_pack, self.U_f.bias, self.nu, a_flat,
+                mask.flatten(1)
+            )
+            return ctx, alpha
         # 1-channel conv via unfold+GEMM (MIOpen naive-kernel bypass, see
         # encoder.conv_1ch) — this runs once per decode timestep.
         from .encoder import conv_1ch
@@ -87,7 +97,8 @@ class CoverageAttention(nn.Module):
         coverage: Tensor,   # [B, H', W'] running sum of alphas
         mask: Tensor,       # [B, H', W'] feature-grid mask
         w_cov: Optional[Tensor] = None,
+        w_pack=None,
     ) -> Tuple[Tensor, Tensor]:
         """Returns (context[B,D], alpha[B,L])."""
         q = self.W_a(s_hat)                                 # [B,A]
-        return self.attend(q, a_flat, ua, coverage, mask, w_cov)
+        return self.attend(q, a_flat, ua, coverage, mask, w_cov, w_pack)

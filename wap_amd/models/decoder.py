@@ -109,12 +109,14 @@ class Parser(nn.Module):
         mask: Tensor,         # [B, H', W']
         fused: bool = False,
         w_cov: Optional[Tensor] = None,
+        w_pack=None,
     ) -> Tuple[Tensor, Tensor, Tensor, Tensor]:
         """One decode step -> (logits, s_t, c_t, alpha)."""
         gru1 = self.gru1.step_fused if fused else self.gru1
         gru2 = self.gru2.step_fused if fused else self.gru2
         s_hat = gru1(y_prev_emb, s_prev)
-        c_t, alpha = self.attention(s_hat, a_flat, ua, coverage, mask, w_cov)
+        c_t, alpha = self.attention(s_hat, a_flat, ua, coverage, mask,
+                                    w_cov, w_pack)
         s_t = gru2(c_t, s_hat)
         logits = self.readout(s_t, c_t, y_prev_emb)
         return logits, s_t, c_t, alpha
@@ -158,13 +160,16 @@ class Parser(nn.Module):
         w_qh = torch.cat([att.W_a.weight, self.gru2.w_h], dim=0)  # [A+3n, n]
         b_qh = torch.cat([self.gru2.b_h.new_zeros(A), self.gru2.b_h])
         w_cov = att.merged_coverage_weight()
+        from ..ops import functional as opf
+        w_pack = (opf.pack_coverage_weight(w_cov)
+                  if opf.can_fuse_cov_score(coverage, w_cov, ua) else None)
         ss, cs = [], []
         for t in range(T):
             gh1 = torch.nn.functional.linear(s, self.gru1.w_h, self.gru1.b_h)
             s_hat = ops.gru_gates(gx1_all[:, t], gh1, s)
             qh = torch.nn.functional.linear(s_hat, w_qh, b_qh)
             c_t, alpha = att.attend(qh[:, :A], a_flat, ua, coverage, mask,
-                                    w_cov)
+                                    w_cov, w_pack)
             gx2 = torch.nn.functional.linear(c_t, self.gru2.w_x, self.gru2.b_x)
             s = ops.gru_gates(gx2, qh[:, A:], s_hat)
             coverage = coverage + alpha.view(b, hp, wp)

@@ -141,6 +141,93 @@ def coverage_attention(
 
 
 # --------------------------------------------------------------------------
+# fully-fused coverage-attention score: 11x11 conv over the coverage map
+# (merged weight, bias), + q + ua, tanh, nu-dot — ONE kernel fwd, one bwd
+# (csrc/fattn.hip), replacing im2col + GEMM + bias-add + attn_score per
+# decoder timestep.
+# --------------------------------------------------------------------------
+
+def pack_coverage_weight(w_merged: Tensor) -> Tuple[Tensor, Tensor]:
+    """Prepare the merged [A,1,KH,KW] coverage weight for the fused kernel:
+    (wpad [A,KP] bf16, wT [KP,A] bf16), KP = KH*KW rounded up to 32.  Both
+    DETACHED — grads route to w_merged directly through _FusedCovScore.
+    Call once per sequence / per decode problem, not per timestep."""
+    A, _, KH, KW = w_merged.shape
+    KK = KH * KW
+    KP = ((KK + 31) // 32) * 32
+    with torch.no_grad():
+        wpad = torch.nn.functional.pad(
+            w_merged.view(A, KK).float(), (0, KP - KK)
+        ).to(torch.bfloat16).contiguous()
+        wT = wpad.t().contiguous()
+    return wpad, wT
+
+
+class _FusedCovScore(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, cov: Tensor, w_merged: Tensor, wpad: Tensor, wT: Tensor,
+                bias: Tensor, q: Tensor, ua: Tensor, nu: Tensor) -> Tensor:
+        # w_merged is never read here (wpad/wT are its packed form) — it is
+        # an input so autograd routes the f32 weight grad to it and
+        # accumulates across the T timestep calls in f32.
+        from . import ext
+        A, _, KH, KW = w_merged.shape
+        nu_f = _up(nu).contiguous()
+        cov_f = _up(cov).contiguous()
+        e, t = ext.require().fattn_score_fwd(
+            cov_f, wpad, q.to(ua.dtype).contiguous(), ua.contiguous(),
+            nu_f, bias.float().contiguous(), KH, KW)
+        ctx.save_for_backward(t, nu_f, wT, cov_f)
+        ctx.meta = (KH, KW, cov.shape[1], cov.shape[2],
+                    q.dtype, w_merged.dtype, bias.dtype, nu.dtype, cov.dtype)
+        return e
+
+    @staticmethod
+    def backward(ctx, de: Tensor):
+        t, nu_f, wT, cov_f = ctx.saved_tensors
+        KH, KW, Hp, Wp, qd, wd, bd, nud, cd = ctx.meta
+        from . import ext
+        dua, dq, dnu, dcov = ext.require().fattn_score_bwd(
+            de.float().contiguous(), t, nu_f, wT, Hp, Wp, KH, KW)
+        # weight grad reduces over B*L: one batched GEMM on the host beats
+        # 64k atomics per tile in-kernel.  dW[a,kk] = sum_{b,l} g * cols
+        A = t.shape[2]
+        (cols,) = ext.require().im2col_1ch(cov_f, KH, KW, 1, KH // 2)
+        dW = torch.bmm(dua.transpose(1, 2),
+                       cols.to(dua.dtype).transpose(1, 2))
+        dW = dW.float().sum(0).view(A, 1, KH, KW)
+        dbias = dq.sum(0)
+        return (dcov.to(cd), dW.to(wd), None, None, dbias.to(bd), dq.to(qd),
+                dua, dnu.to(nud))
+
+
+def can_fuse_cov_score(cov: Tensor, w_merged: Tensor, ua: Tensor) -> bool:
+    A, _, KH, KW = w_merged.shape
+    return (
+        use_hip(ua)
+        and not op_eager("fattn")
+        and ua.dtype == torch.bfloat16
+        and KH == KW and KH % 2 == 1 and KH * KW <= 128
+        and A % 64 == 0 and A <= 512
+        and ua.shape[1] == cov.shape[1] * cov.shape[2]
+    )
+
+
+def fused_cov_attention(
+    q: Tensor, ua: Tensor, cov: Tensor, w_merged: Tensor, w_pack, bias: Tensor,
+    nu: Tensor, a: Tensor, mask: Tensor,
+) -> Tuple[Tensor, Tensor]:
+    """Coverage conv + score in one launch, then masked softmax + context.
+    Caller must have checked :func:`can_fuse_cov_score`.  ``w_pack`` is
+    ``pack_coverage_weight(w_merged)`` (hoisted per sequence) or None."""
+    if w_pack is None:
+        w_pack = pack_coverage_weight(w_merged)
+    e = _FusedCovScore.apply(cov, w_merged, w_pack[0], w_pack[1], bias, q,
+                             ua, nu)
+    return masked_softmax_context(e, a, mask)
+
+
+# --------------------------------------------------------------------------
 # batched 1-channel unfold (im2col) — torch's unfold launches one kernel per
 # image; the HIP op is one launch for the whole batch (fwd and bwd).
 # --------------------------------------------------------------------------

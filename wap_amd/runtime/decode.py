@@ -50,7 +50,7 @@ def _beam_step_(model, S: Dict[str, Tensor], b: int, k: int, fused: bool) -> Non
     hp, wp = S["coverage"].shape[1], S["coverage"].shape[2]
     logits, s_new, _, alpha = m.step(
         S["emb"], S["s"], S["a_flat"], S["ua"], S["coverage"], S["mask"],
-        fused=fused, w_cov=S["w_cov"],
+        fused=fused, w_cov=S["w_cov"], w_pack=(S["wpad"], S["wT"]),
     )
     logprob = torch.log_softmax(logits.float(), dim=-1).view(b, k, -1)
     K = logprob.shape[-1]
@@ -95,12 +95,18 @@ class BeamState:
         mask_bk = _expand_for_beams(mask, beam)
         a_flat_bk = _expand_for_beams(a_flat, beam).contiguous()
         s0 = model.parser.init_state(a_flat_bk, mask_bk.flatten(1))
+        # merged coverage-conv weight (+ its packed form for the fused
+        # kernel), built once per decode problem — derived from live
+        # weights; reset_from refreshes them so cached graphs never replay
+        # a stale copy after training steps
+        w_cov = (model.parser.attention.merged_coverage_weight()
+                 .detach().contiguous())
+        from ..ops.functional import pack_coverage_weight
+        wpad, wT = pack_coverage_weight(w_cov)
         self.S: Dict[str, Tensor] = {
-            # merged coverage-conv weight, built once per decode problem
-            # (derived from live weights; reset_from refreshes it so cached
-            # graphs never replay a stale copy after training steps)
-            "w_cov": model.parser.attention.merged_coverage_weight()
-                     .detach().contiguous(),
+            "w_cov": w_cov,
+            "wpad": wpad,
+            "wT": wT,
             "a_flat": a_flat_bk,
             "ua": _expand_for_beams(ua, beam).contiguous(),
             "mask": mask_bk.contiguous(),
@@ -131,7 +137,8 @@ class BeamState:
         the same shape: copy the per-problem tensors in, reset the mutable
         state.  The hipGraph stays valid (same storage)."""
         S, O = self.S, other.S
-        for n in ("a_flat", "ua", "mask", "s", "coverage", "emb", "w_cov"):
+        for n in ("a_flat", "ua", "mask", "s", "coverage", "emb", "w_cov",
+                  "wpad", "wT"):
             S[n].copy_(O[n])
         S["scores"].fill_(NEG_INF)
         S["scores"][:, 0] = 0.0
