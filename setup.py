@@ -22,7 +22,7 @@ setup(
         CUDAExtension(
             name="wap_amd._wap_hip",
             sources=["wap_amd/csrc/wap_ext.hip", "wap_amd/csrc/conv_mfma.hip",
-                     "wap_amd/csrc/fattn.hip"],
+                     "wap_amd/csrc/fattn.hip", "wap_amd/csrc/convbwd.hip"],
             extra_compile_args={
                 "cxx": ["-O3"],
                 "nvcc": ["-O3", "-std=c++17"],

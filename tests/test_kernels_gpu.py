@@ -357,3 +357,62 @@ def test_fused_cov_score_matches_unfused_hip():
         opf.op_eager = old
     assert _rel_err(ctx1, ctx2) < 3e-2
     assert _rel_err(al1, al2) < 5e-2
+
+
+@pytest.mark.parametrize("B,C,H,W,K,R", [
+    (4, 48, 16, 32, 96, 1),    # dense-block 1x1 (generic-C path)
+    (4, 432, 8, 16, 96, 1),    # 1x1, C%96==0 (bk96 path)
+    (4, 96, 16, 32, 24, 3),    # bottleneck 3x3 (halo path)
+    (2, 684, 8, 32, 96, 1),    # widest block-3 input
+    (2, 60, 7, 13, 24, 3),     # odd spatial, non-96 C, generic 3x3
+])
+@pytest.mark.parametrize("training", [True, False])
+def test_fused_bnreluconv_full_autograd(B, C, H, W, K, R, training):
+    """Hand-written stats + fwd + dgrad/wrw/BN-backward vs the fp32 eager
+    composition (train-mode BatchNorm -> ReLU -> conv2d), all grads."""
+    from wap_amd.ops.fused_conv import _BNReluConv
+    torch.manual_seed(21)
+    dev = "cuda"
+    x0 = torch.randn(B, C, H, W, device=dev)
+    w0 = torch.randn(K, C, R, R, device=dev) * (C * R * R) ** -0.5
+    gamma0 = torch.rand(C, device=dev) + 0.5
+    beta0 = torch.randn(C, device=dev) * 0.3
+    rm0 = torch.randn(C, device=dev) * 0.1
+    rv0 = torch.rand(C, device=dev) + 0.5
+
+    # ---- fp32 eager oracle
+    x = x0.clone().requires_grad_()
+    w = w0.clone().requires_grad_()
+    gamma = gamma0.clone().requires_grad_()
+    beta = beta0.clone().requires_grad_()
+    rm, rv = rm0.clone(), rv0.clone()
+    xn = torch.nn.functional.batch_norm(x, rm, rv, gamma, beta,
+                                        training=training, momentum=0.1,
+                                        eps=1e-5)
+    out_ref = torch.nn.functional.conv2d(torch.relu(xn), w, padding=R // 2)
+    torch.manual_seed(23)
+    gout = torch.randn_like(out_ref)
+    out_ref.backward(gout)
+
+    # ---- fused bf16 path
+    xh = x0.clone().to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last).requires_grad_()
+    wh = w0.clone().to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last).requires_grad_()
+    gammah = gamma0.clone().requires_grad_()
+    betah = beta0.clone().requires_grad_()
+    rmh, rvh = rm0.clone(), rv0.clone()
+    out_hip = _BNReluConv.apply(xh, wh, gammah, betah, rmh, rvh, 1e-5, 0.1,
+                                training)
+    out_hip.backward(gout.to(torch.bfloat16)
+                     .contiguous(memory_format=torch.channels_last))
+
+    assert _rel_err(out_hip, out_ref) < 5e-2, ("out", _rel_err(out_hip, out_ref))
+    assert _rel_err(rmh, rm) < 2e-2
+    assert _rel_err(rvh, rv) < 2e-2
+    pairs = [("dx", xh.grad, x.grad), ("dw", wh.grad, w.grad),
+             ("dgamma", gammah.grad, gamma.grad),
+             ("dbeta", betah.grad, beta.grad)]
+    for name, h, r in pairs:
+        err = _rel_err(h, r)
+        assert err < 8e-2, (name, err)

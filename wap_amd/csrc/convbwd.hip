@@ -1,0 +1,493 @@
+// Backward + batch-stats kernels for the fused BN-ReLU-conv bottleneck path
+// (gfx950 / CDNA4, NHWC bf16, f32 accumulation).
+//
+// Forward (conv_mfma.hip) computes out = conv(relu(bn(x)), w) with the BN
+// affine fused into the input load.  Round-1 left the op default-OFF
+// because the autograd wrapper paid for torch var_mean batch stats, an f32
+// recompute of relu(bn(x)) in backward and aten.convolution_backward
+// (PERF.md "End-to-end ... 254 vs 407").  These kernels remove all of
+// that:
+//
+//   ch_stats       per-channel sum/sumsq over B*H*W (one reduction kernel)
+//   bn_finalize    scale/shift/mean/invstd from the sums + running-stat
+//                  update, one tiny [C] kernel (runs inside captured
+//                  graphs, so replayed steps update running stats too)
+//   bnconv_dgrad   dxn = conv_transpose(g_out, w) as an MFMA implicit
+//                  GEMM, fused with the relu mask (recomputed from x) and
+//                  the BN reductions dbeta = sum g, dgamma = sum g*xhat
+//   conv_wrw       dw[k,r,s,c] = sum_p g_out[p,k] * relu(bn(x))[p+D,c],
+//                  MFMA over LDS-transposed tiles, split-K over pixel
+//                  chunks with f32 atomics
+//   bn_bwd_dx      the final elementwise train-mode BN backward
+//
+// Weight-layout conventions match conv_mfma.hip: w physical [K][R][S][C]
+// (channels_last), x/out physical [M=B*H*W][C].  The dgrad kernel reads a
+// transposed, padded weight wT [R*S][Cp][Kp] prepared by the wrapper.
+
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+
+#include <vector>
+
+#define WAVE 64
+
+typedef __hip_bfloat16 cb_bf16;
+typedef __attribute__((ext_vector_type(8))) short cb_bf16x8;
+typedef __attribute__((ext_vector_type(4))) float cb_f32x4;
+
+static __device__ __forceinline__ float cb_b2f(cb_bf16 v) {
+  return __bfloat162float(v);
+}
+static __device__ __forceinline__ cb_bf16 cb_f2b(float v) {
+  return __float2bfloat16(v);
+}
+
+// ------------------------------------------------------- channel stats
+// x [M, C] bf16 (NHWC flat).  C <= 768 (3 register slots at 256 threads).
+__global__ void ch_stats_kernel(
+    const cb_bf16* __restrict__ x, float* __restrict__ sums,
+    float* __restrict__ sumsq, long M, int C) {
+  const long per = (M + gridDim.x - 1) / gridDim.x;
+  const long p0 = (long)blockIdx.x * per;
+  const long p1 = min(p0 + per, M);
+  const int tid = threadIdx.x;
+  float s[3] = {0.f, 0.f, 0.f}, q[3] = {0.f, 0.f, 0.f};
+  for (long p = p0; p < p1; ++p) {
+    const cb_bf16* row = x + p * C;
+    int j = 0;
+    for (int c = tid; c < C; c += blockDim.x, ++j) {
+      const float v = cb_b2f(row[c]);
+      s[j] += v;
+      q[j] += v * v;
+    }
+  }
+  int j = 0;
+  for (int c = tid; c < C; c += blockDim.x, ++j) {
+    atomicAdd(sums + c, s[j]);
+    atomicAdd(sumsq + c, q[j]);
+  }
+}
+
+// ------------------------------------------------------- bn finalize
+// One thread per channel: batch stats -> (scale, shift, mean, invstd),
+// running-stat update in place (training mode).
+__global__ void bn_finalize_kernel(
+    const float* __restrict__ sums, const float* __restrict__ sumsq,
+    const float* __restrict__ gamma, const float* __restrict__ beta,
+    float* __restrict__ running_mean, float* __restrict__ running_var,
+    float* __restrict__ scale, float* __restrict__ shift,
+    float* __restrict__ mean_out, float* __restrict__ invstd_out,
+    float n, float eps, float momentum, int training, int C) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float mean, var;
+  if (training) {
+    mean = sums[c] / n;
+    var = fmaxf(sumsq[c] / n - mean * mean, 0.f);
+    if (momentum > 0.f) {
+      const float unb = var * n / fmaxf(n - 1.f, 1.f);
+      running_mean[c] = (1.f - momentum) * running_mean[c] + momentum * mean;
+      running_var[c] = (1.f - momentum) * running_var[c] + momentum * unb;
+    }
+  } else {
+    mean = running_mean[c];
+    var = running_var[c];
+  }
+  const float inv = rsqrtf(var + eps);
+  const float sc = gamma[c] * inv;
+  scale[c] = sc;
+  shift[c] = beta[c] - mean * sc;
+  mean_out[c] = mean;
+  invstd_out[c] = inv;
+}
+
+// ------------------------------------------------------------- dgrad
+// g[p,c] = relu'(bn(x)[p,c]) * sum_{k,r,s} g_out[p - D(r,s), k] * w[k,r,s,c]
+// plus dbeta[c] += g, dgamma[c] += g * xhat.  One (128-pixel x 64-channel)
+// tile per block; K padded to Kp (mult 32), C padded to Cp (mult 64) in wT.
+#define DG_BM 128
+#define DG_BN 64
+
+__global__ __launch_bounds__(256, 2)
+void bnconv_dgrad_kernel(
+    const cb_bf16* __restrict__ gout, const cb_bf16* __restrict__ wT,
+    const cb_bf16* __restrict__ x, const float* __restrict__ scale,
+    const float* __restrict__ shift, const float* __restrict__ mean,
+    const float* __restrict__ invstd, cb_bf16* __restrict__ g,
+    float* __restrict__ dgamma, float* __restrict__ dbeta,
+    int Bn, int H, int W, int C, int K, int Kp, int R, int S) {
+  __shared__ cb_bf16 lG[DG_BM][40];        // one 32-wide K-chunk, +8 pad
+  const int HW = H * W;
+  const long M = (long)Bn * HW;
+  const long m0 = (long)blockIdx.x * DG_BM;
+  const int c0 = blockIdx.y * DG_BN;
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wv = tid >> 6;
+  const int PR = R / 2, PS = S / 2;
+  const int Cp = ((C + DG_BN - 1) / DG_BN) * DG_BN;
+
+  cb_f32x4 acc[2][4];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  for (int r = 0; r < R; ++r) {
+    for (int s = 0; s < S; ++s) {
+      for (int kc = 0; kc < Kp; kc += 32) {
+        // stage g_out[p - D(r,s), kc..kc+32) for the tile's 128 pixels
+#pragma unroll
+        for (int i = 0; i < DG_BM * 4 / 256; ++i) {
+          const int flat = tid * (DG_BM * 4 / 256) + i;
+          const int arow = flat >> 2;           // pixel within tile
+          const int apc = flat & 3;             // 8-element piece
+          const long p = m0 + arow;
+          bool inb = false;
+          long base = 0;
+          if (p < M) {
+            const int b = (int)(p / HW);
+            const int rem = (int)(p % HW);
+            const int y = rem / W - (r - PR);
+            const int xw = rem % W - (s - PS);
+            if (y >= 0 && y < H && xw >= 0 && xw < W) {
+              inb = true;
+              base = (((long)b * H + y) * W + xw) * K + kc + apc * 8;
+            }
+          }
+#pragma unroll
+          for (int e = 0; e < 8; ++e) {
+            const int k = kc + apc * 8 + e;
+            lG[arow][apc * 8 + e] =
+                (inb && k < K) ? gout[base + e] : cb_f2b(0.f);
+          }
+        }
+        __syncthreads();
+        const long wtap = ((long)(r * S + s) * Cp) * Kp;
+#pragma unroll
+        for (int mf = 0; mf < 2; ++mf) {
+          const cb_bf16x8 afrag = *(const cb_bf16x8*)
+              &lG[wv * 32 + mf * 16 + (lane & 15)][(lane >> 4) * 8];
+#pragma unroll
+          for (int nf = 0; nf < 4; ++nf) {
+            const cb_bf16x8 bfrag = *(const cb_bf16x8*)
+                &wT[wtap + (long)(c0 + nf * 16 + (lane & 15)) * Kp + kc +
+                    (lane >> 4) * 8];
+            acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afrag, bfrag, acc[mf][nf], 0, 0, 0);
+          }
+        }
+        __syncthreads();
+      }
+    }
+  }
+
+  // epilogue: relu mask from x, store g, reduce dbeta/dgamma
+#pragma unroll
+  for (int nf = 0; nf < 4; ++nf) {
+    const int c = c0 + nf * 16 + (lane & 15);
+    const bool cv = c < C;
+    const float sc = cv ? scale[c] : 0.f;
+    const float sh = cv ? shift[c] : 0.f;
+    const float mn = cv ? mean[c] : 0.f;
+    const float is = cv ? invstd[c] : 0.f;
+    float db = 0.f, dg = 0.f;
+#pragma unroll
+    for (int mf = 0; mf < 2; ++mf) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const long p = m0 + wv * 32 + mf * 16 + (lane >> 4) * 4 + reg;
+        if (cv && p < M) {
+          const float xv = cb_b2f(x[p * C + c]);
+          const float gg =
+              (xv * sc + sh) > 0.f ? acc[mf][nf][reg] : 0.f;
+          g[p * C + c] = cb_f2b(gg);
+          db += gg;
+          dg += gg * (xv - mn) * is;
+        }
+      }
+    }
+    db += __shfl_xor(db, 16, WAVE);
+    db += __shfl_xor(db, 32, WAVE);
+    dg += __shfl_xor(dg, 16, WAVE);
+    dg += __shfl_xor(dg, 32, WAVE);
+    if ((lane >> 4) == 0 && cv) {
+      atomicAdd(dbeta + c, db);
+      atomicAdd(dgamma + c, dg);
+    }
+  }
+}
+
+// --------------------------------------------------------------- wrw
+// dw[k, (r,s), c] = sum_p g_out[p, k] * relu(bn(x))[p + D(r,s), c].
+// Block: all K rows (<= 96) x 64 channels, split-K over pixel chunks with
+// f32 atomics into dw.  g_out and the recomputed xn are LDS-transposed so
+// both MFMA operands are contiguous 16-byte reads.
+#define WR_PIX 2048
+
+__global__ __launch_bounds__(256, 2)
+void conv_wrw_kernel(
+    const cb_bf16* __restrict__ gout, const cb_bf16* __restrict__ x,
+    const float* __restrict__ scale, const float* __restrict__ shift,
+    float* __restrict__ dw,
+    int Bn, int H, int W, int C, int K, int R, int S) {
+  __shared__ cb_bf16 lGT[96][40];          // [k][pixel-in-chunk], +8 pad
+  __shared__ cb_bf16 lX[64][40];           // [c][pixel-in-chunk], +8 pad
+  __shared__ float lsc[64], lsh[64];
+
+  const int HW = H * W;
+  const long M = (long)Bn * HW;
+  const long pc0 = (long)blockIdx.x * WR_PIX;
+  const int c0 = blockIdx.y * 64;
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wv = tid >> 6;
+  const int PR = R / 2, PS = S / 2;
+  const int MF = (K + 15) / 16;            // <= 6
+
+  if (tid < 64) {
+    const int c = c0 + tid;
+    lsc[tid] = c < C ? scale[c] : 0.f;
+    lsh[tid] = c < C ? shift[c] : 0.f;
+  }
+
+  for (int tap = 0; tap < R * S; ++tap) {
+    const int r = tap / S, s = tap % S;
+    cb_f32x4 acc[6];
+#pragma unroll
+    for (int i = 0; i < 6; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+
+    for (long p0 = pc0; p0 < min(pc0 + WR_PIX, M); p0 += 32) {
+      __syncthreads();
+      // stage g_out transposed: read [pp][k] coalesced, write lGT[k][pp]
+      for (int flat = tid; flat < 32 * K; flat += blockDim.x) {
+        const int pp = flat / K, k = flat % K;
+        const long p = p0 + pp;
+        lGT[k][pp] = p < M ? gout[p * K + k] : cb_f2b(0.f);
+      }
+      // stage xn (= relu(bn(x)) at the tap-shifted pixel) transposed
+      for (int flat = tid; flat < 32 * 64; flat += blockDim.x) {
+        const int pp = flat >> 6, cc = flat & 63;
+        const long p = p0 + pp;
+        float v = 0.f;
+        if (p < M && c0 + cc < C) {
+          const int b = (int)(p / HW);
+          const int rem = (int)(p % HW);
+          const int y = rem / W + (r - PR);
+          const int xw = rem % W + (s - PS);
+          if (y >= 0 && y < H && xw >= 0 && xw < W) {
+            v = cb_b2f(x[(((long)b * H + y) * W + xw) * C + c0 + cc]) *
+                    lsc[cc] + lsh[cc];
+            v = fmaxf(v, 0.f);
+          }
+        }
+        lX[cc][pp] = cb_f2b(v);
+      }
+      __syncthreads();
+      const cb_bf16x8 bfrag = *(const cb_bf16x8*)
+          &lX[wv * 16 + (lane & 15)][(lane >> 4) * 8];
+#pragma unroll
+      for (int mf = 0; mf < 6; ++mf) {
+        if (mf >= MF) break;
+        const cb_bf16x8 afrag = *(const cb_bf16x8*)
+            &lGT[mf * 16 + (lane & 15)][(lane >> 4) * 8];
+        acc[mf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag, bfrag, acc[mf], 0, 0, 0);
+      }
+    }
+    // accumulate this tap's partial into dw (f32 atomics)
+    const int c = c0 + wv * 16 + (lane & 15);
+#pragma unroll
+    for (int mf = 0; mf < 6; ++mf) {
+      if (mf >= MF) break;
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int k = mf * 16 + (lane >> 4) * 4 + reg;
+        if (k < K && c < C)
+          atomicAdd(dw + ((long)k * R * S + tap) * C + c, acc[mf][reg]);
+      }
+    }
+    __syncthreads();
+  }
+}
+
+// ---------------------------------------------------------- bn_bwd_dx
+// dx = gamma*invstd * (g - dbeta/n - xhat * dgamma/n)   (training)
+// dx = gamma*invstd * g                                 (eval)
+__global__ void bn_bwd_dx_kernel(
+    const cb_bf16* __restrict__ g, const cb_bf16* __restrict__ x,
+    const float* __restrict__ gamma, const float* __restrict__ invstd,
+    const float* __restrict__ mean, const float* __restrict__ dgamma,
+    const float* __restrict__ dbeta, float inv_n, cb_bf16* __restrict__ dx,
+    long total, int C, int training) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= total) return;
+  const int c = (int)(i % C);
+  const float is = invstd[c];
+  const float gs = gamma[c] * is;
+  const float gv = cb_b2f(g[i]);
+  float v;
+  if (training) {
+    const float xhat = (cb_b2f(x[i]) - mean[c]) * is;
+    v = gs * (gv - dbeta[c] * inv_n - xhat * dgamma[c] * inv_n);
+  } else {
+    v = gs * gv;
+  }
+  dx[i] = cb_f2b(v);
+}
+
+// ================================================================ host
+namespace {
+inline hipStream_t cb_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+__global__ void cb_zero_kernel(float* __restrict__ p, long n) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) p[i] = 0.f;
+}
+void cb_zero(torch::Tensor& t) {
+  const long n = t.numel();
+  hipLaunchKernelGGL(cb_zero_kernel, dim3((n + 255) / 256), dim3(256), 0,
+                     cb_stream(), (float*)t.data_ptr(), n);
+}
+#define CB_OK(what) do { \
+  hipError_t err = hipGetLastError(); \
+  TORCH_CHECK(err == hipSuccess, what ": ", hipGetErrorString(err)); \
+} while (0)
+}  // namespace
+
+std::vector<torch::Tensor> ch_stats(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "ch_stats: x must be channels_last");
+  const int C = x.size(1);
+  TORCH_CHECK(C <= 768, "ch_stats: C <= 768");
+  const long M = x.numel() / C;
+  auto sums = torch::empty({C}, x.options().dtype(at::kFloat));
+  auto sumsq = torch::empty({C}, x.options().dtype(at::kFloat));
+  cb_zero(sums);
+  cb_zero(sumsq);
+  const int blocks = (int)std::min((M + 127) / 128, (long)512);
+  hipLaunchKernelGGL(ch_stats_kernel, dim3(blocks), dim3(256), 0,
+                     cb_stream(), (const cb_bf16*)x.data_ptr(),
+                     (float*)sums.data_ptr(), (float*)sumsq.data_ptr(), M, C);
+  CB_OK("ch_stats");
+  return {sums, sumsq};
+}
+
+std::vector<torch::Tensor> bn_finalize(
+    torch::Tensor sums, torch::Tensor sumsq, torch::Tensor gamma,
+    torch::Tensor beta, torch::Tensor running_mean, torch::Tensor running_var,
+    double n, double eps, double momentum, bool training) {
+  const int C = gamma.size(0);
+  auto opts = gamma.options().dtype(at::kFloat);
+  auto scale = torch::empty({C}, opts);
+  auto shift = torch::empty({C}, opts);
+  auto mean = torch::empty({C}, opts);
+  auto invstd = torch::empty({C}, opts);
+  TORCH_CHECK(gamma.scalar_type() == at::kFloat &&
+              beta.scalar_type() == at::kFloat &&
+              running_mean.scalar_type() == at::kFloat &&
+              running_var.scalar_type() == at::kFloat,
+              "bn_finalize: f32 BN params expected");
+  hipLaunchKernelGGL(bn_finalize_kernel, dim3((C + 255) / 256), dim3(256), 0,
+                     cb_stream(),
+                     (const float*)sums.data_ptr(),
+                     (const float*)sumsq.data_ptr(),
+                     (const float*)gamma.data_ptr(),
+                     (const float*)beta.data_ptr(),
+                     (float*)running_mean.data_ptr(),
+                     (float*)running_var.data_ptr(),
+                     (float*)scale.data_ptr(), (float*)shift.data_ptr(),
+                     (float*)mean.data_ptr(), (float*)invstd.data_ptr(),
+                     (float)n, (float)eps, (float)momentum,
+                     training ? 1 : 0, C);
+  CB_OK("bn_finalize");
+  return {scale, shift, mean, invstd};
+}
+
+std::vector<torch::Tensor> bnconv_dgrad(
+    torch::Tensor gout, torch::Tensor wT, torch::Tensor x,
+    torch::Tensor scale, torch::Tensor shift, torch::Tensor mean,
+    torch::Tensor invstd, int64_t R, int64_t S) {
+  TORCH_CHECK(gout.is_cuda() && gout.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(gout.is_contiguous(at::MemoryFormat::ChannelsLast));
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast));
+  TORCH_CHECK(wT.is_contiguous() && wT.scalar_type() == at::kBFloat16);
+  const int Bn = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  const int K = gout.size(1);
+  const int Kp = wT.size(2);
+  const int Cp = wT.size(1);
+  TORCH_CHECK(wT.size(0) == R * S && Kp % 32 == 0 && Kp >= K);
+  TORCH_CHECK(Cp % DG_BN == 0 && Cp >= C);
+  auto g = at::empty_like(x);
+  auto dgamma = torch::empty({C}, scale.options());
+  auto dbeta = torch::empty({C}, scale.options());
+  cb_zero(dgamma);
+  cb_zero(dbeta);
+  const long M = (long)Bn * H * W;
+  const dim3 grid((M + DG_BM - 1) / DG_BM, (C + DG_BN - 1) / DG_BN);
+  hipLaunchKernelGGL(bnconv_dgrad_kernel, grid, dim3(256), 0, cb_stream(),
+                     (const cb_bf16*)gout.data_ptr(),
+                     (const cb_bf16*)wT.data_ptr(),
+                     (const cb_bf16*)x.data_ptr(),
+                     (const float*)scale.data_ptr(),
+                     (const float*)shift.data_ptr(),
+                     (const float*)mean.data_ptr(),
+                     (const float*)invstd.data_ptr(),
+                     (cb_bf16*)g.data_ptr(), (float*)dgamma.data_ptr(),
+                     (float*)dbeta.data_ptr(),
+                     Bn, H, W, C, K, Kp, (int)R, (int)S);
+  CB_OK("bnconv_dgrad");
+  return {g, dgamma, dbeta};
+}
+
+torch::Tensor conv_wrw(
+    torch::Tensor gout, torch::Tensor x, torch::Tensor scale,
+    torch::Tensor shift, int64_t R, int64_t S) {
+  TORCH_CHECK(gout.is_contiguous(at::MemoryFormat::ChannelsLast));
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast));
+  const int Bn = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  const int K = gout.size(1);
+  TORCH_CHECK(K <= 96, "conv_wrw: K <= 96");
+  auto dw = torch::empty({K, (long)R * S, C},
+                         x.options().dtype(at::kFloat));
+  cb_zero(dw);
+  const long M = (long)Bn * H * W;
+  const dim3 grid((M + WR_PIX - 1) / WR_PIX, (C + 63) / 64);
+  hipLaunchKernelGGL(conv_wrw_kernel, grid, dim3(256), 0, cb_stream(),
+                     (const cb_bf16*)gout.data_ptr(),
+                     (const cb_bf16*)x.data_ptr(),
+                     (const float*)scale.data_ptr(),
+                     (const float*)shift.data_ptr(),
+                     (float*)dw.data_ptr(),
+                     Bn, H, W, C, K, (int)R, (int)S);
+  CB_OK("conv_wrw");
+  return dw;  // [K, R*S, C] f32
+}
+
+torch::Tensor bn_bwd_dx(
+    torch::Tensor g, torch::Tensor x, torch::Tensor gamma,
+    torch::Tensor invstd, torch::Tensor mean, torch::Tensor dgamma,
+    torch::Tensor dbeta, double n, bool training) {
+  TORCH_CHECK(g.is_contiguous(at::MemoryFormat::ChannelsLast));
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast));
+  const int C = x.size(1);
+  auto dx = at::empty_like(x);
+  const long total = x.numel();
+  hipLaunchKernelGGL(bn_bwd_dx_kernel,
+                     dim3((total + 255) / 256), dim3(256), 0, cb_stream(),
+                     (const cb_bf16*)g.data_ptr(),
+                     (const cb_bf16*)x.data_ptr(),
+                     (const float*)gamma.data_ptr(),
+                     (const float*)invstd.data_ptr(),
+                     (const float*)mean.data_ptr(),
+                     (const float*)dgamma.data_ptr(),
+                     (const float*)dbeta.data_ptr(),
+                     (float)(1.0 / n), (cb_bf16*)dx.data_ptr(),
+                     total, C, training ? 1 : 0);
+  CB_OK("bn_bwd_dx");
+  return dx;
+}

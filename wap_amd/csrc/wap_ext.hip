@@ -864,9 +864,32 @@ std::vector<torch::Tensor> fattn_score_bwd(
     torch::Tensor de, torch::Tensor tcache, torch::Tensor nu,
     torch::Tensor wTpad, int64_t Hp, int64_t Wp, int64_t KH, int64_t KW);
 
+// fused bottleneck backward + stats (convbwd.hip)
+std::vector<torch::Tensor> ch_stats(torch::Tensor x);
+std::vector<torch::Tensor> bn_finalize(
+    torch::Tensor sums, torch::Tensor sumsq, torch::Tensor gamma,
+    torch::Tensor beta, torch::Tensor running_mean, torch::Tensor running_var,
+    double n, double eps, double momentum, bool training);
+std::vector<torch::Tensor> bnconv_dgrad(
+    torch::Tensor gout, torch::Tensor wT, torch::Tensor x,
+    torch::Tensor scale, torch::Tensor shift, torch::Tensor mean,
+    torch::Tensor invstd, int64_t R, int64_t S);
+torch::Tensor conv_wrw(
+    torch::Tensor gout, torch::Tensor x, torch::Tensor scale,
+    torch::Tensor shift, int64_t R, int64_t S);
+torch::Tensor bn_bwd_dx(
+    torch::Tensor g, torch::Tensor x, torch::Tensor gamma,
+    torch::Tensor invstd, torch::Tensor mean, torch::Tensor dgamma,
+    torch::Tensor dbeta, double n, bool training);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fattn_score_fwd", &fattn_score_fwd);
   m.def("fattn_score_bwd", &fattn_score_bwd);
+  m.def("ch_stats", &ch_stats);
+  m.def("bn_finalize", &bn_finalize);
+  m.def("bnconv_dgrad", &bnconv_dgrad);
+  m.def("conv_wrw", &conv_wrw);
+  m.def("bn_bwd_dx", &bn_bwd_dx);
   m.def("bnrelu_conv_fwd", &bnrelu_conv_fwd);
   m.def("bnrelu_conv_fwd_fp8", &bnrelu_conv_fwd_fp8);
   m.def("attention_score_fwd", &attention_score_fwd);

@@ -23,72 +23,74 @@ from torch import Tensor
 from . import use_hip
 
 
-def _bn_stats(x: Tensor, running_mean: Tensor, running_var: Tensor,
-              training: bool, momentum: float):
-    with torch.no_grad():  # stats used numerically; batch-stat dependence
-        if training:       # is handled analytically in _BNReluConv.backward
-            var, mean = torch.var_mean(x.float(), dim=(0, 2, 3), unbiased=False)
-            n = x.numel() / x.shape[1]
-            running_mean.mul_(1 - momentum).add_(mean, alpha=momentum)
-            running_var.mul_(1 - momentum).add_(var * n / max(n - 1, 1),
-                                                alpha=momentum)
-            return mean, var
-        return running_mean.float(), running_var.float()
-
-
 class _BNReluConv(torch.autograd.Function):
+    """relu(bn(x)) -> conv, everything hand-written CDNA4 kernels:
+
+    forward : ch_stats + bn_finalize (one reduction + one [C] kernel,
+              replacing torch var_mean and the running-stat update ops)
+              -> bnrelu_conv_fwd (MFMA implicit GEMM, BN+ReLU in the load)
+    backward: bnconv_dgrad (MFMA conv-transpose fused with the relu mask
+              and the dbeta/dgamma reductions) + conv_wrw (MFMA, split-K
+              atomics) + bn_bwd_dx (one elementwise kernel)
+
+    No aten.convolution_backward, no f32 recompute, no autocast casts —
+    the round-1 wrapper overhead that kept this op default-off (PERF.md).
+    """
+
     @staticmethod
     def forward(ctx, x: Tensor, weight: Tensor, gamma: Tensor, beta: Tensor,
-                mean: Tensor, var: Tensor, eps: float, training: bool):
+                running_mean: Tensor, running_var: Tensor, eps: float,
+                momentum: float, training: bool):
         from . import ext
-        invstd = torch.rsqrt(var + eps)
-        scale = (gamma.float() * invstd).contiguous()
-        shift = (beta.float() - mean * scale).contiguous()
-        out = ext.require().bnrelu_conv_fwd(x, weight, scale, shift, True)
-        ctx.save_for_backward(x, weight, mean, invstd, gamma, scale, shift)
+        lib = ext.require()
+        C = x.shape[1]
+        if training:
+            sums, sumsq = lib.ch_stats(x)
+            n = x.numel() / C
+        else:
+            sums = sumsq = torch.empty(0)
+            n = 1.0
+        scale, shift, mean, invstd = lib.bn_finalize(
+            sums, sumsq, gamma.float(), beta.float(), running_mean,
+            running_var, float(n), eps, momentum if training else 0.0,
+            training)
+        out = lib.bnrelu_conv_fwd(x, weight, scale, shift, True)
+        ctx.save_for_backward(x, weight, gamma, scale, shift, mean, invstd)
         ctx.training = training
+        ctx.n = n
         return out
 
     @staticmethod
     def backward(ctx, grad_out: Tensor):
-        x, weight, mean, invstd, gamma, scale, shift = ctx.saved_tensors
-        grad_out = grad_out.contiguous(
-            memory_format=torch.channels_last)
-        xf = x.float()
-        sc = scale.view(1, -1, 1, 1)
-        sh = shift.view(1, -1, 1, 1)
-        xn = torch.relu(xf * sc + sh).to(x.dtype) \
-            .contiguous(memory_format=torch.channels_last)
-        pad = weight.shape[2] // 2
-        dxn, dw, _ = torch.ops.aten.convolution_backward(
-            grad_out, xn, weight, None, [1, 1], [pad, pad], [1, 1], False,
-            [0, 0], 1, [True, True, False])
-        # relu mask
-        g = dxn.float() * (xn > 0).float()
-        # y = gamma*xhat + beta; xhat = (x - mean)*invstd
-        xhat = (xf - mean.view(1, -1, 1, 1)) * invstd.view(1, -1, 1, 1)
-        dbeta = g.sum(dim=(0, 2, 3))
-        dgamma = (g * xhat).sum(dim=(0, 2, 3))
-        gs = (gamma.float() * invstd).view(1, -1, 1, 1)
-        if ctx.training:
-            n = g.numel() / g.shape[1]
-            dx = gs * (g - dbeta.view(1, -1, 1, 1) / n
-                       - xhat * dgamma.view(1, -1, 1, 1) / n)
-        else:
-            dx = gs * g
-        dx = dx.to(x.dtype).contiguous(memory_format=torch.channels_last)
-        return (dx, dw.to(weight.dtype), dgamma.to(gamma.dtype),
-                dbeta.to(gamma.dtype), None, None, None, None)
+        from . import ext
+        lib = ext.require()
+        x, weight, gamma, scale, shift, mean, invstd = ctx.saved_tensors
+        grad_out = grad_out.contiguous(memory_format=torch.channels_last)
+        K, C, R, S = weight.shape
+        Kp = ((K + 31) // 32) * 32
+        Cp = ((C + 63) // 64) * 64
+        # wT [R*S, Cp, Kp] from the channels_last weight (physical KRSC):
+        # one pad kernel (materializes the permuted view zero-padded)
+        wphys = weight.permute(0, 2, 3, 1).reshape(K, R * S, C)
+        wT = torch.nn.functional.pad(wphys.permute(1, 2, 0),
+                                     (0, Kp - K, 0, Cp - C)).contiguous()
+        g, dgamma, dbeta = lib.bnconv_dgrad(
+            grad_out, wT, x, scale, shift, mean, invstd, R, S)
+        dw = lib.conv_wrw(grad_out, x, scale, shift, R, S)  # [K,R*S,C] f32
+        dx = lib.bn_bwd_dx(g, x, gamma.float(), invstd, mean, dgamma, dbeta,
+                           ctx.n, ctx.training)
+        # physical [K,RS,C] -> logical [K,C,R,S] grad for the cl weight
+        dwl = dw.view(K, R, S, C).permute(0, 3, 1, 2).to(weight.dtype)
+        return (dx, dwl, dgamma.to(gamma.dtype), dbeta.to(gamma.dtype),
+                None, None, None, None, None)
 
 
 def can_fuse(x: Tensor, weight: Tensor) -> bool:
     import os
-    # Default OFF: the kernel itself beats MIOpen 2-3x on the 1x1 shapes
-    # (tools/bench_conv.py) but the training wrapper's batch-stat and
-    # f32-recompute overhead currently costs more than the kernel saves
-    # (profiles/: var_mean + casts ~16% of the step).  Opt in with
-    # WAP_FUSED_CONV=1x1 (1x1 convs) or =all.
-    mode = os.environ.get("WAP_FUSED_CONV", "0")
+    # Default ON (auto): forward beats MIOpen on every bottleneck shape and
+    # the backward is now hand-written too.  WAP_FUSED_CONV=0 disables,
+    # =1x1 restricts to the 1x1 convs (A/B debugging).
+    mode = os.environ.get("WAP_FUSED_CONV", "auto")
     if mode == "0":
         return False
     if mode == "1x1" and weight.shape[2] != 1:
@@ -97,6 +99,7 @@ def can_fuse(x: Tensor, weight: Tensor) -> bool:
         use_hip(x)
         and x.dtype == torch.bfloat16
         and x.is_contiguous(memory_format=torch.channels_last)
+        and x.shape[1] <= 768
         and weight.shape[0] <= 96
         and weight.shape[2] == weight.shape[3]
         and weight.shape[2] % 2 == 1
@@ -105,11 +108,9 @@ def can_fuse(x: Tensor, weight: Tensor) -> bool:
 
 def bnrelu_conv(x: Tensor, weight: Tensor, bn: torch.nn.BatchNorm2d) -> Tensor:
     """relu(bn(x)) -> conv (stride 1, same padding), fused on HIP."""
-    mean, var = _bn_stats(x, bn.running_mean, bn.running_var,
-                          bn.training, bn.momentum)
     w = weight.to(torch.bfloat16).contiguous(memory_format=torch.channels_last)
-    return _BNReluConv.apply(x, w, bn.weight, bn.bias, mean, var,
-                             bn.eps, bn.training)
+    return _BNReluConv.apply(x, w, bn.weight, bn.bias, bn.running_mean,
+                             bn.running_var, bn.eps, bn.momentum, bn.training)
 
 
 # ----------------------------------------------------------------- fp8 path
