@@ -15,9 +15,9 @@
 //   bnconv_dgrad   dxn = conv_transpose(g_out, w) as an MFMA implicit
 //                  GEMM, fused with the relu mask (recomputed from x) and
 //                  the BN reductions dbeta = sum g, dgamma = sum g*xhat
-//   conv_wrw       dw[k,r,s,c] = sum_p g_out[p,k] * relu(bn(x))[p+D,c],
-//                  MFMA over LDS-transposed tiles, split-K over pixel
-//                  chunks with f32 atomics
+//   bnrelu_apply   materialize xn = relu(bn(x)) once for the backward-side
+//                  wrw GEMM (rocBLAS 1x1 / MIOpen 3x3 — measured faster
+//                  than a hand LDS-transposed wrw kernel, see PERF.md)
 //   bn_bwd_dx      the final elementwise train-mode BN backward
 //
 // Weight-layout conventions match conv_mfma.hip: w physical [K][R][S][C]
@@ -45,29 +45,69 @@ static __device__ __forceinline__ cb_bf16 cb_f2b(float v) {
 }
 
 // ------------------------------------------------------- channel stats
-// x [M, C] bf16 (NHWC flat).  C <= 768 (3 register slots at 256 threads).
+// x [M, C] bf16 (NHWC flat).  Each thread owns one 8-channel group
+// (16-byte vector loads) and a stripe of pixels; per-thread f32
+// accumulators, one atomicAdd per channel per block at the end.
 __global__ void ch_stats_kernel(
     const cb_bf16* __restrict__ x, float* __restrict__ sums,
     float* __restrict__ sumsq, long M, int C) {
+  const int ncb = (C + 7) / 8;               // 8-channel groups
+  const int tid = threadIdx.x;
+  const int cb = tid % ncb;
+  const int psub = tid / ncb;
+  const int nps = blockDim.x / ncb;          // pixel lanes per group
+  if (psub >= nps && nps > 0) return;
   const long per = (M + gridDim.x - 1) / gridDim.x;
   const long p0 = (long)blockIdx.x * per;
   const long p1 = min(p0 + per, M);
-  const int tid = threadIdx.x;
-  float s[3] = {0.f, 0.f, 0.f}, q[3] = {0.f, 0.f, 0.f};
-  for (long p = p0; p < p1; ++p) {
-    const cb_bf16* row = x + p * C;
-    int j = 0;
-    for (int c = tid; c < C; c += blockDim.x, ++j) {
-      const float v = cb_b2f(row[c]);
-      s[j] += v;
-      q[j] += v * v;
+  const int c0 = cb * 8;
+  const bool full = c0 + 8 <= C;
+  float s[8] = {0.f}, q[8] = {0.f};
+  for (long p = p0 + psub; p < p1; p += nps) {
+    const cb_bf16* row = x + p * C + c0;
+    if (full) {
+      const cb_bf16x8 v8 = *(const cb_bf16x8*)row;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const short sv = v8[e];
+        const float v = cb_b2f(*(const cb_bf16*)&sv);
+        s[e] += v;
+        q[e] += v * v;
+      }
+    } else {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        if (c0 + e < C) {
+          const float v = cb_b2f(row[e]);
+          s[e] += v;
+          q[e] += v * v;
+        }
+      }
     }
   }
-  int j = 0;
-  for (int c = tid; c < C; c += blockDim.x, ++j) {
-    atomicAdd(sums + c, s[j]);
-    atomicAdd(sumsq + c, q[j]);
+#pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    if (c0 + e < C) {
+      atomicAdd(sums + c0 + e, s[e]);
+      atomicAdd(sumsq + c0 + e, q[e]);
+    }
   }
+}
+
+// ------------------------------------------------- xn materialization
+// xn = relu(x*scale + shift), bf16 NHWC — backward-side input for the
+// library wrw GEMM (the forward never materializes it; backward reads x
+// a couple of times anyway, so one cheap write is the fast wrw route —
+// the hand-written LDS-transposed wrw kernel measured 1.4 ms/call and
+// was replaced by this + rocBLAS/MIOpen, see PERF.md).
+__global__ void bnrelu_apply_kernel(
+    const cb_bf16* __restrict__ x, const float* __restrict__ scale,
+    const float* __restrict__ shift, cb_bf16* __restrict__ xn,
+    long total, int C) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= total) return;
+  const int c = (int)(i % C);
+  xn[i] = cb_f2b(fmaxf(cb_b2f(x[i]) * scale[c] + shift[c], 0.f));
 }
 
 // ------------------------------------------------------- bn finalize
@@ -220,99 +260,6 @@ void bnconv_dgrad_kernel(
   }
 }
 
-// --------------------------------------------------------------- wrw
-// dw[k, (r,s), c] = sum_p g_out[p, k] * relu(bn(x))[p + D(r,s), c].
-// Block: all K rows (<= 96) x 64 channels, split-K over pixel chunks with
-// f32 atomics into dw.  g_out and the recomputed xn are LDS-transposed so
-// both MFMA operands are contiguous 16-byte reads.
-#define WR_PIX 2048
-
-__global__ __launch_bounds__(256, 2)
-void conv_wrw_kernel(
-    const cb_bf16* __restrict__ gout, const cb_bf16* __restrict__ x,
-    const float* __restrict__ scale, const float* __restrict__ shift,
-    float* __restrict__ dw,
-    int Bn, int H, int W, int C, int K, int R, int S) {
-  __shared__ cb_bf16 lGT[96][40];          // [k][pixel-in-chunk], +8 pad
-  __shared__ cb_bf16 lX[64][40];           // [c][pixel-in-chunk], +8 pad
-  __shared__ float lsc[64], lsh[64];
-
-  const int HW = H * W;
-  const long M = (long)Bn * HW;
-  const long pc0 = (long)blockIdx.x * WR_PIX;
-  const int c0 = blockIdx.y * 64;
-  const int tid = threadIdx.x;
-  const int lane = tid & (WAVE - 1);
-  const int wv = tid >> 6;
-  const int PR = R / 2, PS = S / 2;
-  const int MF = (K + 15) / 16;            // <= 6
-
-  if (tid < 64) {
-    const int c = c0 + tid;
-    lsc[tid] = c < C ? scale[c] : 0.f;
-    lsh[tid] = c < C ? shift[c] : 0.f;
-  }
-
-  for (int tap = 0; tap < R * S; ++tap) {
-    const int r = tap / S, s = tap % S;
-    cb_f32x4 acc[6];
-#pragma unroll
-    for (int i = 0; i < 6; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
-
-    for (long p0 = pc0; p0 < min(pc0 + WR_PIX, M); p0 += 32) {
-      __syncthreads();
-      // stage g_out transposed: read [pp][k] coalesced, write lGT[k][pp]
-      for (int flat = tid; flat < 32 * K; flat += blockDim.x) {
-        const int pp = flat / K, k = flat % K;
-        const long p = p0 + pp;
-        lGT[k][pp] = p < M ? gout[p * K + k] : cb_f2b(0.f);
-      }
-      // stage xn (= relu(bn(x)) at the tap-shifted pixel) transposed
-      for (int flat = tid; flat < 32 * 64; flat += blockDim.x) {
-        const int pp = flat >> 6, cc = flat & 63;
-        const long p = p0 + pp;
-        float v = 0.f;
-        if (p < M && c0 + cc < C) {
-          const int b = (int)(p / HW);
-          const int rem = (int)(p % HW);
-          const int y = rem / W + (r - PR);
-          const int xw = rem % W + (s - PS);
-          if (y >= 0 && y < H && xw >= 0 && xw < W) {
-            v = cb_b2f(x[(((long)b * H + y) * W + xw) * C + c0 + cc]) *
-                    lsc[cc] + lsh[cc];
-            v = fmaxf(v, 0.f);
-          }
-        }
-        lX[cc][pp] = cb_f2b(v);
-      }
-      __syncthreads();
-      const cb_bf16x8 bfrag = *(const cb_bf16x8*)
-          &lX[wv * 16 + (lane & 15)][(lane >> 4) * 8];
-#pragma unroll
-      for (int mf = 0; mf < 6; ++mf) {
-        if (mf >= MF) break;
-        const cb_bf16x8 afrag = *(const cb_bf16x8*)
-            &lGT[mf * 16 + (lane & 15)][(lane >> 4) * 8];
-        acc[mf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            afrag, bfrag, acc[mf], 0, 0, 0);
-      }
-    }
-    // accumulate this tap's partial into dw (f32 atomics)
-    const int c = c0 + wv * 16 + (lane & 15);
-#pragma unroll
-    for (int mf = 0; mf < 6; ++mf) {
-      if (mf >= MF) break;
-#pragma unroll
-      for (int reg = 0; reg < 4; ++reg) {
-        const int k = mf * 16 + (lane >> 4) * 4 + reg;
-        if (k < K && c < C)
-          atomicAdd(dw + ((long)k * R * S + tap) * C + c, acc[mf][reg]);
-      }
-    }
-    __syncthreads();
-  }
-}
-
 // ---------------------------------------------------------- bn_bwd_dx
 // dx = gamma*invstd * (g - dbeta/n - xhat * dgamma/n)   (training)
 // dx = gamma*invstd * g                                 (eval)
@@ -444,30 +391,6 @@ std::vector<torch::Tensor> bnconv_dgrad(
   return {g, dgamma, dbeta};
 }
 
-torch::Tensor conv_wrw(
-    torch::Tensor gout, torch::Tensor x, torch::Tensor scale,
-    torch::Tensor shift, int64_t R, int64_t S) {
-  TORCH_CHECK(gout.is_contiguous(at::MemoryFormat::ChannelsLast));
-  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast));
-  const int Bn = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
-  const int K = gout.size(1);
-  TORCH_CHECK(K <= 96, "conv_wrw: K <= 96");
-  auto dw = torch::empty({K, (long)R * S, C},
-                         x.options().dtype(at::kFloat));
-  cb_zero(dw);
-  const long M = (long)Bn * H * W;
-  const dim3 grid((M + WR_PIX - 1) / WR_PIX, (C + 63) / 64);
-  hipLaunchKernelGGL(conv_wrw_kernel, grid, dim3(256), 0, cb_stream(),
-                     (const cb_bf16*)gout.data_ptr(),
-                     (const cb_bf16*)x.data_ptr(),
-                     (const float*)scale.data_ptr(),
-                     (const float*)shift.data_ptr(),
-                     (float*)dw.data_ptr(),
-                     Bn, H, W, C, K, (int)R, (int)S);
-  CB_OK("conv_wrw");
-  return dw;  // [K, R*S, C] f32
-}
-
 torch::Tensor bn_bwd_dx(
     torch::Tensor g, torch::Tensor x, torch::Tensor gamma,
     torch::Tensor invstd, torch::Tensor mean, torch::Tensor dgamma,
@@ -490,4 +413,21 @@ torch::Tensor bn_bwd_dx(
                      total, C, training ? 1 : 0);
   CB_OK("bn_bwd_dx");
   return dx;
+}
+
+torch::Tensor bnrelu_apply(
+    torch::Tensor x, torch::Tensor scale, torch::Tensor shift) {
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast));
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16);
+  const int C = x.size(1);
+  auto xn = at::empty_like(x);
+  const long total = x.numel();
+  hipLaunchKernelGGL(bnrelu_apply_kernel,
+                     dim3((total + 255) / 256), dim3(256), 0, cb_stream(),
+                     (const cb_bf16*)x.data_ptr(),
+                     (const float*)scale.data_ptr(),
+                     (const float*)shift.data_ptr(),
+                     (cb_bf16*)xn.data_ptr(), total, C);
+  CB_OK("bnrelu_apply");
+  return xn;
 }

@@ -874,9 +874,8 @@ std::vector<torch::Tensor> bnconv_dgrad(
     torch::Tensor gout, torch::Tensor wT, torch::Tensor x,
     torch::Tensor scale, torch::Tensor shift, torch::Tensor mean,
     torch::Tensor invstd, int64_t R, int64_t S);
-torch::Tensor conv_wrw(
-    torch::Tensor gout, torch::Tensor x, torch::Tensor scale,
-    torch::Tensor shift, int64_t R, int64_t S);
+torch::Tensor bnrelu_apply(
+    torch::Tensor x, torch::Tensor scale, torch::Tensor shift);
 torch::Tensor bn_bwd_dx(
     torch::Tensor g, torch::Tensor x, torch::Tensor gamma,
     torch::Tensor invstd, torch::Tensor mean, torch::Tensor dgamma,
@@ -888,7 +887,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ch_stats", &ch_stats);
   m.def("bn_finalize", &bn_finalize);
   m.def("bnconv_dgrad", &bnconv_dgrad);
-  m.def("conv_wrw", &conv_wrw);
+  m.def("bnrelu_apply", &bnrelu_apply);
   m.def("bn_bwd_dx", &bn_bwd_dx);
   m.def("bnrelu_conv_fwd", &bnrelu_conv_fwd);
   m.def("bnrelu_conv_fwd_fp8", &bnrelu_conv_fwd_fp8);

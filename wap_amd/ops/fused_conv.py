@@ -76,13 +76,26 @@ class _BNReluConv(torch.autograd.Function):
                                      (0, Kp - K, 0, Cp - C)).contiguous()
         g, dgamma, dbeta = lib.bnconv_dgrad(
             grad_out, wT, x, scale, shift, mean, invstd, R, S)
-        dw = lib.conv_wrw(grad_out, x, scale, shift, R, S)  # [K,R*S,C] f32
+        # wrw: materialize xn once (one elementwise kernel), then it is a
+        # plain GEMM — rocBLAS for 1x1, MIOpen igemm for 3x3.  The
+        # hand-written LDS-transposed wrw kernel measured 1.4 ms/call vs
+        # ~20 us here (PERF.md): reduction over the fast (channel-adjacent)
+        # pixel dim wants a library GEMM, not a per-tile transpose.
+        xn = lib.bnrelu_apply(x, scale, shift)
+        if R == 1:
+            M = xn.numel() // C
+            g2 = grad_out.permute(0, 2, 3, 1).reshape(M, K)   # free (cl)
+            x2 = xn.permute(0, 2, 3, 1).reshape(M, C)
+            dwl = (g2.t() @ x2).view(K, C, 1, 1)
+        else:
+            pad = R // 2
+            _, dwl, _ = torch.ops.aten.convolution_backward(
+                grad_out, xn, weight, None, [1, 1], [pad, pad], [1, 1],
+                False, [0, 0], 1, [False, True, False])
         dx = lib.bn_bwd_dx(g, x, gamma.float(), invstd, mean, dgamma, dbeta,
                            ctx.n, ctx.training)
-        # physical [K,RS,C] -> logical [K,C,R,S] grad for the cl weight
-        dwl = dw.view(K, R, S, C).permute(0, 3, 1, 2).to(weight.dtype)
-        return (dx, dwl, dgamma.to(gamma.dtype), dbeta.to(gamma.dtype),
-                None, None, None, None, None)
+        return (dx, dwl.to(weight.dtype), dgamma.to(gamma.dtype),
+                dbeta.to(gamma.dtype), None, None, None, None, None)
 
 
 def can_fuse(x: Tensor, weight: Tensor) -> bool:
