@@ -45,25 +45,25 @@ static __device__ __forceinline__ cb_bf16 cb_f2b(float v) {
 }
 
 // ------------------------------------------------------- channel stats
-// x [M, C] bf16 (NHWC flat).  Each thread owns one 8-channel group
-// (16-byte vector loads) and a stripe of pixels; per-thread f32
-// accumulators, one atomicAdd per channel per block at the end.
+// x [M, C] bf16 (NHWC flat).  Grid (pixel-chunks, 8-channel groups):
+// every thread of a block accumulates the SAME 8 channels over a pixel
+// stripe (16-byte vector loads), then a block-wide shuffle/LDS reduction
+// leaves ONE atomicAdd per channel per block (<=64 contending writers per
+// address — a first version with per-thread atomics serialized ~20k adds
+// per address on small-C layers and ran 10x slower than the BN it
+// replaced).
 __global__ void ch_stats_kernel(
     const cb_bf16* __restrict__ x, float* __restrict__ sums,
     float* __restrict__ sumsq, long M, int C) {
-  const int ncb = (C + 7) / 8;               // 8-channel groups
+  __shared__ float red[4][16];
   const int tid = threadIdx.x;
-  const int cb = tid % ncb;
-  const int psub = tid / ncb;
-  const int nps = blockDim.x / ncb;          // pixel lanes per group
-  if (psub >= nps && nps > 0) return;
-  const long per = (M + gridDim.x - 1) / gridDim.x;
-  const long p0 = (long)blockIdx.x * per;
-  const long p1 = min(p0 + per, M);
-  const int c0 = cb * 8;
+  const int lane = tid & (WAVE - 1);
+  const int wv = tid >> 6;
+  const int c0 = blockIdx.y * 8;
   const bool full = c0 + 8 <= C;
+  const long step = (long)gridDim.x * blockDim.x;
   float s[8] = {0.f}, q[8] = {0.f};
-  for (long p = p0 + psub; p < p1; p += nps) {
+  for (long p = (long)blockIdx.x * blockDim.x + tid; p < M; p += step) {
     const cb_bf16* row = x + p * C + c0;
     if (full) {
       const cb_bf16x8 v8 = *(const cb_bf16x8*)row;
@@ -85,11 +85,29 @@ __global__ void ch_stats_kernel(
       }
     }
   }
+  // wave reduce each of the 16 accumulators, then across the 4 waves
 #pragma unroll
   for (int e = 0; e < 8; ++e) {
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      s[e] += __shfl_down(s[e], off, WAVE);
+      q[e] += __shfl_down(q[e], off, WAVE);
+    }
+  }
+  if (lane == 0) {
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      red[wv][e] = s[e];
+      red[wv][8 + e] = q[e];
+    }
+  }
+  __syncthreads();
+  if (tid < 16) {
+    const float v = red[0][tid] + red[1][tid] + red[2][tid] + red[3][tid];
+    const int e = tid & 7;
     if (c0 + e < C) {
-      atomicAdd(sums + c0 + e, s[e]);
-      atomicAdd(sumsq + c0 + e, q[e]);
+      if (tid < 8) atomicAdd(sums + c0 + e, v);
+      else atomicAdd(sumsq + c0 + e, v);
     }
   }
 }
@@ -316,8 +334,9 @@ std::vector<torch::Tensor> ch_stats(torch::Tensor x) {
   auto sumsq = torch::empty({C}, x.options().dtype(at::kFloat));
   cb_zero(sums);
   cb_zero(sumsq);
-  const int blocks = (int)std::min((M + 127) / 128, (long)512);
-  hipLaunchKernelGGL(ch_stats_kernel, dim3(blocks), dim3(256), 0,
+  const int pch = (int)std::min((M + 1023) / 1024, (long)64);
+  const dim3 grid(pch, (C + 7) / 8);
+  hipLaunchKernelGGL(ch_stats_kernel, grid, dim3(256), 0,
                      cb_stream(), (const cb_bf16*)x.data_ptr(),
                      (float*)sums.data_ptr(), (float*)sumsq.data_ptr(), M, C);
   CB_OK("ch_stats");
