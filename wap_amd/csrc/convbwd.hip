@@ -176,7 +176,12 @@ void bnconv_dgrad_kernel(
     const float* __restrict__ invstd, cb_bf16* __restrict__ g,
     float* __restrict__ dgamma, float* __restrict__ dbeta,
     int Bn, int H, int W, int C, int K, int Kp, int R, int S) {
-  __shared__ cb_bf16 lG[DG_BM][40];        // one 32-wide K-chunk, +8 pad
+  // Whole reduction axis (all taps x all K-chunks, R*S*Kp <= 288) staged
+  // ONCE: two barriers for the entire block (a per-(tap,chunk) staging
+  // loop measured ~70us/call = 3x MIOpen igemm_bwd).  Each thread owns a
+  // fixed pixel row of the tile, so the p -> (b,y,x) division happens
+  // once, not once per staged piece.
+  __shared__ cb_bf16 lG[DG_BM][296];       // up to 9 taps x 32 K, +8 pad
   const int HW = H * W;
   const long M = (long)Bn * HW;
   const long m0 = (long)blockIdx.x * DG_BM;
@@ -186,6 +191,7 @@ void bnconv_dgrad_kernel(
   const int wv = tid >> 6;
   const int PR = R / 2, PS = S / 2;
   const int Cp = ((C + DG_BN - 1) / DG_BN) * DG_BN;
+  const int TK = R * S * Kp;               // staged row length (<= 288)
 
   cb_f32x4 acc[2][4];
 #pragma unroll
@@ -193,54 +199,70 @@ void bnconv_dgrad_kernel(
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  for (int r = 0; r < R; ++r) {
-    for (int s = 0; s < S; ++s) {
-      for (int kc = 0; kc < Kp; kc += 32) {
-        // stage g_out[p - D(r,s), kc..kc+32) for the tile's 128 pixels
+  {
+    // threads t and t+128 share pixel row t%128, staging half its pieces
+    const int arow = tid & (DG_BM - 1);
+    const int half = tid >> 7;             // 0 or 1
+    const long p = m0 + arow;
+    int b = 0, y0 = 0, x0 = 0;
+    const bool pv = p < M;
+    if (pv) {
+      b = (int)(p / HW);
+      const int rem = (int)(p % HW);
+      y0 = rem / W;
+      x0 = rem % W;
+    }
+    const int npieces = TK / 8;
+    for (int pc = half; pc < npieces; pc += 2) {
+      const int koff = pc * 8;             // offset into the staged row
+      const int tap = koff / Kp;
+      const int kk = koff % Kp;            // multiple of 8 within Kp
+      cb_bf16 vals[8];
+      bool done = false;
+      if (pv) {
+        const int y = y0 - (tap / S - PR);
+        const int xw = x0 - (tap % S - PS);
+        if (y >= 0 && y < H && xw >= 0 && xw < W) {
+          const long base = (((long)b * H + y) * W + xw) * K + kk;
+          if (kk + 8 <= K) {
+            *(cb_bf16x8*)vals = *(const cb_bf16x8*)(gout + base);
+            done = true;
+          } else {
 #pragma unroll
-        for (int i = 0; i < DG_BM * 4 / 256; ++i) {
-          const int flat = tid * (DG_BM * 4 / 256) + i;
-          const int arow = flat >> 2;           // pixel within tile
-          const int apc = flat & 3;             // 8-element piece
-          const long p = m0 + arow;
-          bool inb = false;
-          long base = 0;
-          if (p < M) {
-            const int b = (int)(p / HW);
-            const int rem = (int)(p % HW);
-            const int y = rem / W - (r - PR);
-            const int xw = rem % W - (s - PS);
-            if (y >= 0 && y < H && xw >= 0 && xw < W) {
-              inb = true;
-              base = (((long)b * H + y) * W + xw) * K + kc + apc * 8;
-            }
-          }
-#pragma unroll
-          for (int e = 0; e < 8; ++e) {
-            const int k = kc + apc * 8 + e;
-            lG[arow][apc * 8 + e] =
-                (inb && k < K) ? gout[base + e] : cb_f2b(0.f);
-          }
-        }
-        __syncthreads();
-        const long wtap = ((long)(r * S + s) * Cp) * Kp;
-#pragma unroll
-        for (int mf = 0; mf < 2; ++mf) {
-          const cb_bf16x8 afrag = *(const cb_bf16x8*)
-              &lG[wv * 32 + mf * 16 + (lane & 15)][(lane >> 4) * 8];
-#pragma unroll
-          for (int nf = 0; nf < 4; ++nf) {
-            const cb_bf16x8 bfrag = *(const cb_bf16x8*)
-                &wT[wtap + (long)(c0 + nf * 16 + (lane & 15)) * Kp + kc +
-                    (lane >> 4) * 8];
-            acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                afrag, bfrag, acc[mf][nf], 0, 0, 0);
+            for (int e = 0; e < 8; ++e)
+              vals[e] = kk + e < K ? gout[base + e] : cb_f2b(0.f);
+            done = true;
           }
         }
-        __syncthreads();
+      }
+      if (!done) {
+#pragma unroll
+        for (int e = 0; e < 8; ++e) vals[e] = cb_f2b(0.f);
+      }
+      *(cb_bf16x8*)&lG[arow][koff] = *(cb_bf16x8*)vals;
+    }
+  }
+  __syncthreads();
+  for (int tap = 0; tap < R * S; ++tap) {
+    const long wtap = ((long)tap * Cp) * Kp;
+    for (int kc = 0; kc < Kp; kc += 32) {
+#pragma unroll
+      for (int mf = 0; mf < 2; ++mf) {
+        const cb_bf16x8 afrag = *(const cb_bf16x8*)
+            &lG[wv * 32 + mf * 16 + (lane & 15)][tap * Kp + kc +
+                                                 (lane >> 4) * 8];
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf) {
+          const cb_bf16x8 bfrag = *(const cb_bf16x8*)
+              &wT[wtap + (long)(c0 + nf * 16 + (lane & 15)) * Kp + kc +
+                  (lane >> 4) * 8];
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag, bfrag, acc[mf][nf], 0, 0, 0);
+        }
       }
     }
   }
+  __syncthreads();
 
   // epilogue: relu mask from x, store g, reduce dbeta/dgamma
 #pragma unroll
@@ -388,6 +410,7 @@ std::vector<torch::Tensor> bnconv_dgrad(
   const int Cp = wT.size(1);
   TORCH_CHECK(wT.size(0) == R * S && Kp % 32 == 0 && Kp >= K);
   TORCH_CHECK(Cp % DG_BN == 0 && Cp >= C);
+  TORCH_CHECK(R * S * Kp <= 288, "bnconv_dgrad: R*S*Kp <= 288");
   auto g = at::empty_like(x);
   auto dgamma = torch::empty({C}, scale.options());
   auto dbeta = torch::empty({C}, scale.options());

@@ -82,16 +82,12 @@ class _BNReluConv(torch.autograd.Function):
         # ~20 us here (PERF.md): reduction over the fast (channel-adjacent)
         # pixel dim wants a library GEMM, not a per-tile transpose.
         xn = lib.bnrelu_apply(x, scale, shift)
-        if R == 1:
-            M = xn.numel() // C
-            g2 = grad_out.permute(0, 2, 3, 1).reshape(M, K)   # free (cl)
-            x2 = xn.permute(0, 2, 3, 1).reshape(M, C)
-            dwl = (g2.t() @ x2).view(K, C, 1, 1)
-        else:
-            pad = R // 2
-            _, dwl, _ = torch.ops.aten.convolution_backward(
-                grad_out, xn, weight, None, [1, 1], [pad, pad], [1, 1],
-                False, [0, 0], 1, [False, True, False])
+        pad = R // 2
+        # MIOpen igemm wrw (~19us at these shapes; rocBLAS on the same
+        # tall-skinny [K,M]x[M,C] TN GEMM measured 166us — no split-K)
+        _, dwl, _ = torch.ops.aten.convolution_backward(
+            grad_out, xn, weight, None, [1, 1], [pad, pad], [1, 1],
+            False, [0, 0], 1, [False, True, False])
         dx = lib.bn_bwd_dx(g, x, gamma.float(), invstd, mean, dgamma, dbeta,
                            ctx.n, ctx.training)
         return (dx, dwl.to(weight.dtype), dgamma.to(gamma.dtype),
