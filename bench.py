@@ -115,7 +115,7 @@ def main() -> None:
             out = model(x, x_mask, y, y_mask)
         out["loss"].backward()
         ddp.finish_gradient_sync()
-        torch.nn.utils.clip_grad_norm_(model.parameters(), 100.0)
+        ddp.clip_grad_norm_(100.0)
         opt.step()
         return out["loss"]
 

@@ -416,3 +416,77 @@ def test_fused_bnreluconv_full_autograd(B, C, H, W, K, R, training):
     for name, h, r in pairs:
         err = _rel_err(h, r)
         assert err < 8e-2, (name, err)
+
+
+@pytest.mark.parametrize("B,C,H,W,K", [
+    (4, 432, 16, 32, 216),   # transition 1 shape class
+    (2, 600, 8, 16, 300),    # transition 2
+])
+@pytest.mark.parametrize("training", [True, False])
+def test_fused_transition_full_autograd(B, C, H, W, K, training):
+    """relu(bn(x)) -> 1x1 conv with K>96 (stats/mask kernels + rocBLAS)
+    vs the fp32 eager composition, all grads."""
+    from wap_amd.ops.fused_conv import _BNReluLinear
+    torch.manual_seed(33)
+    dev = "cuda"
+    x0 = torch.randn(B, C, H, W, device=dev)
+    w0 = torch.randn(K, C, 1, 1, device=dev) * C ** -0.5
+    gamma0 = torch.rand(C, device=dev) + 0.5
+    beta0 = torch.randn(C, device=dev) * 0.3
+    rm0 = torch.randn(C, device=dev) * 0.1
+    rv0 = torch.rand(C, device=dev) + 0.5
+
+    x = x0.clone().requires_grad_()
+    w = w0.clone().requires_grad_()
+    gamma = gamma0.clone().requires_grad_()
+    beta = beta0.clone().requires_grad_()
+    rm, rv = rm0.clone(), rv0.clone()
+    xn = torch.nn.functional.batch_norm(x, rm, rv, gamma, beta,
+                                        training=training, momentum=0.1,
+                                        eps=1e-5)
+    out_ref = torch.nn.functional.conv2d(torch.relu(xn), w)
+    torch.manual_seed(35)
+    gout = torch.randn_like(out_ref)
+    out_ref.backward(gout)
+
+    xh = x0.clone().to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last).requires_grad_()
+    wh = w0.clone().to(torch.bfloat16) \
+        .contiguous(memory_format=torch.channels_last).requires_grad_()
+    gammah = gamma0.clone().requires_grad_()
+    betah = beta0.clone().requires_grad_()
+    rmh, rvh = rm0.clone(), rv0.clone()
+    out_hip = _BNReluLinear.apply(xh, wh, gammah, betah, rmh, rvh, 1e-5,
+                                  0.1, training)
+    out_hip.backward(gout.to(torch.bfloat16)
+                     .contiguous(memory_format=torch.channels_last))
+
+    assert _rel_err(out_hip, out_ref) < 5e-2
+    assert _rel_err(rmh, rm) < 2e-2 and _rel_err(rvh, rv) < 2e-2
+    for name, h, r in [("dx", xh.grad, x.grad), ("dw", wh.grad, w.grad),
+                       ("dgamma", gammah.grad, gamma.grad),
+                       ("dbeta", betah.grad, beta.grad)]:
+        err = _rel_err(h, r)
+        assert err < 8e-2, (name, err)
+
+
+@pytest.mark.parametrize("nbuf", [1, 3])
+def test_bucket_clip_matches_torch(nbuf):
+    """BucketedDDP.clip_grad_norm_ over flat buffers == torch clip."""
+    import torch.nn as nn
+    from wap_amd.parallel import BucketedDDP
+    torch.manual_seed(41)
+    m1 = nn.Sequential(nn.Linear(64, 64), nn.Linear(64, 32)).cuda()
+    m2 = nn.Sequential(nn.Linear(64, 64), nn.Linear(64, 32)).cuda()
+    m2.load_state_dict(m1.state_dict())
+    x = torch.randn(8, 64, device="cuda")
+    for m in (m1, m2):
+        m.zero_grad(set_to_none=False)
+    ddp = BucketedDDP(m2, bucket_bytes=(1 << 30) if nbuf == 1 else 4096)
+    m1(x).square().sum().backward()
+    m2(x).square().sum().backward()
+    n1 = torch.nn.utils.clip_grad_norm_(m1.parameters(), 0.5)
+    n2 = ddp.clip_grad_norm_(0.5)
+    assert torch.allclose(n1, n2, rtol=1e-5)
+    for p1, p2 in zip(m1.parameters(), m2.parameters()):
+        assert torch.allclose(p1.grad, p2.grad, rtol=1e-5, atol=1e-7)

@@ -183,8 +183,7 @@ def main() -> None:
                 out["loss"].backward()
                 ddp.finish_gradient_sync()
                 if cfg.grad_clip > 0:
-                    torch.nn.utils.clip_grad_norm_(model.parameters(),
-                                                   cfg.grad_clip)
+                    ddp.clip_grad_norm_(cfg.grad_clip)
                 optimizer.step()
                 loss_t = out["loss"]
             gstep += 1

@@ -60,7 +60,8 @@ __global__ void ch_stats_kernel(
   const int lane = tid & (WAVE - 1);
   const int wv = tid >> 6;
   const int c0 = blockIdx.y * 8;
-  const bool full = c0 + 8 <= C;
+  // 16-byte vector loads need C %% 8 == 0 (row base p*C stays aligned)
+  const bool full = (c0 + 8 <= C) && (C % 8 == 0);
   const long step = (long)gridDim.x * blockDim.x;
   float s[8] = {0.f}, q[8] = {0.f};
   for (long p = (long)blockIdx.x * blockDim.x + tid; p < M; p += step) {
@@ -121,11 +122,118 @@ __global__ void ch_stats_kernel(
 __global__ void bnrelu_apply_kernel(
     const cb_bf16* __restrict__ x, const float* __restrict__ scale,
     const float* __restrict__ shift, cb_bf16* __restrict__ xn,
-    long total, int C) {
-  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= total) return;
-  const int c = (int)(i % C);
-  xn[i] = cb_f2b(fmaxf(cb_b2f(x[i]) * scale[c] + shift[c], 0.f));
+    long M, int C) {
+  const int c0 = blockIdx.y * 8;
+  const bool full = (c0 + 8 <= C) && (C % 8 == 0);
+  float sc[8], sh[8];
+#pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    sc[e] = c0 + e < C ? scale[c0 + e] : 0.f;
+    sh[e] = c0 + e < C ? shift[c0 + e] : 0.f;
+  }
+  const long step = (long)gridDim.x * blockDim.x;
+  for (long p = (long)blockIdx.x * blockDim.x + threadIdx.x; p < M;
+       p += step) {
+    const cb_bf16* row = x + p * C + c0;
+    cb_bf16* out = xn + p * C + c0;
+    if (full) {
+      cb_bf16 v8[8];
+      *(cb_bf16x8*)v8 = *(const cb_bf16x8*)row;
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        v8[e] = cb_f2b(fmaxf(cb_b2f(v8[e]) * sc[e] + sh[e], 0.f));
+      *(cb_bf16x8*)out = *(cb_bf16x8*)v8;
+    } else {
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        if (c0 + e < C)
+          out[e] = cb_f2b(fmaxf(cb_b2f(row[e]) * sc[e] + sh[e], 0.f));
+    }
+  }
+}
+
+// ------------------------------------------- relu/BN mask + reductions
+// g = relu'(bn(x)) * dxn, dbeta[c] += g, dgamma[c] += g*xhat — the
+// epilogue of bnconv_dgrad as a standalone kernel, for convs whose
+// data-grad is a plain GEMM (transition 1x1, K > 96).
+__global__ void relu_mask_reduce_kernel(
+    const cb_bf16* __restrict__ dxn, const cb_bf16* __restrict__ x,
+    const float* __restrict__ scale, const float* __restrict__ shift,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    cb_bf16* __restrict__ g, float* __restrict__ dgamma,
+    float* __restrict__ dbeta, long M, int C) {
+  __shared__ float red[4][16];
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wv = tid >> 6;
+  const int c0 = blockIdx.y * 8;
+  const bool full = (c0 + 8 <= C) && (C % 8 == 0);
+  float sc[8], sh[8], mn[8], is[8], db[8], dg[8];
+#pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    const int c = c0 + e;
+    const bool cv = c < C;
+    sc[e] = cv ? scale[c] : 0.f;
+    sh[e] = cv ? shift[c] : 0.f;
+    mn[e] = cv ? mean[c] : 0.f;
+    is[e] = cv ? invstd[c] : 0.f;
+    db[e] = dg[e] = 0.f;
+  }
+  const long step = (long)gridDim.x * blockDim.x;
+  for (long p = (long)blockIdx.x * blockDim.x + tid; p < M; p += step) {
+    const cb_bf16* drow = dxn + p * C + c0;
+    const cb_bf16* xrow = x + p * C + c0;
+    cb_bf16* grow = g + p * C + c0;
+    cb_bf16 dv8[8], xv8[8], gv8[8];
+    if (full) {
+      *(cb_bf16x8*)dv8 = *(const cb_bf16x8*)drow;
+      *(cb_bf16x8*)xv8 = *(const cb_bf16x8*)xrow;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const float xv = cb_b2f(xv8[e]);
+        const float gg = (xv * sc[e] + sh[e]) > 0.f ? cb_b2f(dv8[e]) : 0.f;
+        gv8[e] = cb_f2b(gg);
+        db[e] += gg;
+        dg[e] += gg * (xv - mn[e]) * is[e];
+      }
+      *(cb_bf16x8*)grow = *(cb_bf16x8*)gv8;
+    } else {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        if (c0 + e < C) {
+          const float xv = cb_b2f(xrow[e]);
+          const float gg = (xv * sc[e] + sh[e]) > 0.f ? cb_b2f(drow[e]) : 0.f;
+          grow[e] = cb_f2b(gg);
+          db[e] += gg;
+          dg[e] += gg * (xv - mn[e]) * is[e];
+        }
+      }
+    }
+  }
+#pragma unroll
+  for (int e = 0; e < 8; ++e) {
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      db[e] += __shfl_down(db[e], off, WAVE);
+      dg[e] += __shfl_down(dg[e], off, WAVE);
+    }
+  }
+  if (lane == 0) {
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      red[wv][e] = dg[e];
+      red[wv][8 + e] = db[e];
+    }
+  }
+  __syncthreads();
+  if (tid < 16) {
+    const float v = red[0][tid] + red[1][tid] + red[2][tid] + red[3][tid];
+    const int e = tid & 7;
+    if (c0 + e < C) {
+      if (tid < 8) atomicAdd(dgamma + c0 + e, v);
+      else atomicAdd(dbeta + c0 + e, v);
+    }
+  }
 }
 
 // ------------------------------------------------------- bn finalize
@@ -264,38 +372,86 @@ void bnconv_dgrad_kernel(
   }
   __syncthreads();
 
-  // epilogue: relu mask from x, store g, reduce dbeta/dgamma
+  // epilogue: park the raw conv-transpose tile in LDS (reusing lG's
+  // space), then stream it back out row-wise so the x reads and g stores
+  // are 16-byte coalesced — the per-accumulator scatter version issued
+  // 2-byte accesses at a 2*C stride and dominated the kernel.
+  float* lT = (float*)&lG[0][0];           // [DG_BM][DG_BN+4] f32, 34.8 KB
+  const int tstride = DG_BN + 4;
 #pragma unroll
   for (int nf = 0; nf < 4; ++nf) {
-    const int c = c0 + nf * 16 + (lane & 15);
-    const bool cv = c < C;
-    const float sc = cv ? scale[c] : 0.f;
-    const float sh = cv ? shift[c] : 0.f;
-    const float mn = cv ? mean[c] : 0.f;
-    const float is = cv ? invstd[c] : 0.f;
-    float db = 0.f, dg = 0.f;
 #pragma unroll
     for (int mf = 0; mf < 2; ++mf) {
 #pragma unroll
       for (int reg = 0; reg < 4; ++reg) {
-        const long p = m0 + wv * 32 + mf * 16 + (lane >> 4) * 4 + reg;
-        if (cv && p < M) {
-          const float xv = cb_b2f(x[p * C + c]);
-          const float gg =
-              (xv * sc + sh) > 0.f ? acc[mf][nf][reg] : 0.f;
-          g[p * C + c] = cb_f2b(gg);
-          db += gg;
-          dg += gg * (xv - mn) * is;
-        }
+        const int prow = wv * 32 + mf * 16 + (lane >> 4) * 4 + reg;
+        lT[prow * tstride + nf * 16 + (lane & 15)] = acc[mf][nf][reg];
       }
     }
-    db += __shfl_xor(db, 16, WAVE);
-    db += __shfl_xor(db, 32, WAVE);
-    dg += __shfl_xor(dg, 16, WAVE);
-    dg += __shfl_xor(dg, 32, WAVE);
-    if ((lane >> 4) == 0 && cv) {
-      atomicAdd(dbeta + c, db);
-      atomicAdd(dgamma + c, dg);
+  }
+  __syncthreads();
+  {
+    // thread t owns 8-channel piece cp = t>>5 of rows (t&31), (t&31)+32, ...
+    const int cp = tid >> 5;               // 0..7
+    const int cbase = cp * 8;
+    float db[8], dg[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) db[e] = dg[e] = 0.f;
+    float sc[8], sh[8], mn[8], is[8];
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      const int c = c0 + cbase + e;
+      sc[e] = c < C ? scale[c] : 0.f;
+      sh[e] = c < C ? shift[c] : 0.f;
+      mn[e] = c < C ? mean[c] : 0.f;
+      is[e] = c < C ? invstd[c] : 0.f;
+    }
+    const bool cfull = (c0 + cbase + 8 <= C) && (C % 8 == 0);
+    for (int row = tid & 31; row < DG_BM; row += 32) {
+      const long p = m0 + row;
+      if (p >= M) break;
+      cb_bf16 xv8[8], gv8[8];
+      if (cfull)
+        *(cb_bf16x8*)xv8 = *(const cb_bf16x8*)(x + p * C + c0 + cbase);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const int c = c0 + cbase + e;
+        if (c < C) {
+          const float xv = cfull ? cb_b2f(xv8[e]) : cb_b2f(x[p * C + c]);
+          const float gg = (xv * sc[e] + sh[e]) > 0.f
+                               ? lT[row * tstride + cbase + e] : 0.f;
+          gv8[e] = cb_f2b(gg);
+          db[e] += gg;
+          dg[e] += gg * (xv - mn[e]) * is[e];
+        } else {
+          gv8[e] = cb_f2b(0.f);
+        }
+      }
+      if (cfull)
+        *(cb_bf16x8*)(g + p * C + c0 + cbase) = *(cb_bf16x8*)gv8;
+      else
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          if (c0 + cbase + e < C) g[p * C + c0 + cbase + e] = gv8[e];
+    }
+    // reduce across the 32 row-threads sharing this channel piece
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+#pragma unroll
+      for (int off = 1; off < 32; off <<= 1) {
+        db[e] += __shfl_xor(db[e], off, WAVE);
+        dg[e] += __shfl_xor(dg[e], off, WAVE);
+      }
+    }
+    if ((tid & 31) == 0) {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const int c = c0 + cbase + e;
+        if (c < C) {
+          atomicAdd(dbeta + c, db[e]);
+          atomicAdd(dgamma + c, dg[e]);
+        }
+      }
     }
   }
 }
@@ -308,21 +464,47 @@ __global__ void bn_bwd_dx_kernel(
     const float* __restrict__ gamma, const float* __restrict__ invstd,
     const float* __restrict__ mean, const float* __restrict__ dgamma,
     const float* __restrict__ dbeta, float inv_n, cb_bf16* __restrict__ dx,
-    long total, int C, int training) {
-  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i >= total) return;
-  const int c = (int)(i % C);
-  const float is = invstd[c];
-  const float gs = gamma[c] * is;
-  const float gv = cb_b2f(g[i]);
-  float v;
-  if (training) {
-    const float xhat = (cb_b2f(x[i]) - mean[c]) * is;
-    v = gs * (gv - dbeta[c] * inv_n - xhat * dgamma[c] * inv_n);
-  } else {
-    v = gs * gv;
+    long M, int C, int training) {
+  const int c0 = blockIdx.y * 8;
+  const bool full = (c0 + 8 <= C) && (C % 8 == 0);
+  float gs[8], is[8], mn[8], dbn[8], dgn[8];
+#pragma unroll
+  for (int e = 0; e < 8; ++e) {
+    const int c = c0 + e;
+    const bool cv = c < C;
+    is[e] = cv ? invstd[c] : 0.f;
+    gs[e] = cv ? gamma[c] * is[e] : 0.f;
+    mn[e] = cv ? mean[c] : 0.f;
+    dbn[e] = cv && training ? dbeta[c] * inv_n : 0.f;
+    dgn[e] = cv && training ? dgamma[c] * inv_n : 0.f;
   }
-  dx[i] = cb_f2b(v);
+  const long step = (long)gridDim.x * blockDim.x;
+  for (long p = (long)blockIdx.x * blockDim.x + threadIdx.x; p < M;
+       p += step) {
+    const cb_bf16* grow = g + p * C + c0;
+    const cb_bf16* xrow = x + p * C + c0;
+    cb_bf16* orow = dx + p * C + c0;
+    cb_bf16 gv8[8], xv8[8], ov8[8];
+    if (full) {
+      *(cb_bf16x8*)gv8 = *(const cb_bf16x8*)grow;
+      *(cb_bf16x8*)xv8 = *(const cb_bf16x8*)xrow;
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const float xhat = (cb_b2f(xv8[e]) - mn[e]) * is[e];
+        ov8[e] = cb_f2b(gs[e] * (cb_b2f(gv8[e]) - dbn[e] - xhat * dgn[e]));
+      }
+      *(cb_bf16x8*)orow = *(cb_bf16x8*)ov8;
+    } else {
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        if (c0 + e < C) {
+          const float xhat = (cb_b2f(xrow[e]) - mn[e]) * is[e];
+          orow[e] = cb_f2b(gs[e] * (cb_b2f(grow[e]) - dbn[e]
+                                    - xhat * dgn[e]));
+        }
+      }
+    }
+  }
 }
 
 // ================================================================ host
@@ -352,10 +534,10 @@ std::vector<torch::Tensor> ch_stats(torch::Tensor x) {
   const int C = x.size(1);
   TORCH_CHECK(C <= 768, "ch_stats: C <= 768");
   const long M = x.numel() / C;
-  auto sums = torch::empty({C}, x.options().dtype(at::kFloat));
-  auto sumsq = torch::empty({C}, x.options().dtype(at::kFloat));
-  cb_zero(sums);
-  cb_zero(sumsq);
+  auto buf = torch::empty({2, C}, x.options().dtype(at::kFloat));
+  cb_zero(buf);  // one zero launch for both accumulators
+  auto sums = buf.select(0, 0);
+  auto sumsq = buf.select(0, 1);
   const int pch = (int)std::min((M + 1023) / 1024, (long)64);
   const dim3 grid(pch, (C + 7) / 8);
   hipLaunchKernelGGL(ch_stats_kernel, grid, dim3(256), 0,
@@ -412,10 +594,10 @@ std::vector<torch::Tensor> bnconv_dgrad(
   TORCH_CHECK(Cp % DG_BN == 0 && Cp >= C);
   TORCH_CHECK(R * S * Kp <= 288, "bnconv_dgrad: R*S*Kp <= 288");
   auto g = at::empty_like(x);
-  auto dgamma = torch::empty({C}, scale.options());
-  auto dbeta = torch::empty({C}, scale.options());
-  cb_zero(dgamma);
-  cb_zero(dbeta);
+  auto gb = torch::empty({2, C}, scale.options());
+  cb_zero(gb);  // one zero launch for both reductions
+  auto dgamma = gb.select(0, 0);
+  auto dbeta = gb.select(0, 1);
   const long M = (long)Bn * H * W;
   const dim3 grid((M + DG_BM - 1) / DG_BM, (C + DG_BN - 1) / DG_BN);
   hipLaunchKernelGGL(bnconv_dgrad_kernel, grid, dim3(256), 0, cb_stream(),
@@ -441,9 +623,10 @@ torch::Tensor bn_bwd_dx(
   TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast));
   const int C = x.size(1);
   auto dx = at::empty_like(x);
-  const long total = x.numel();
+  const long M = x.numel() / C;
+  const int pch = (int)std::min((M + 255) / 256, (long)128);
   hipLaunchKernelGGL(bn_bwd_dx_kernel,
-                     dim3((total + 255) / 256), dim3(256), 0, cb_stream(),
+                     dim3(pch, (C + 7) / 8), dim3(256), 0, cb_stream(),
                      (const cb_bf16*)g.data_ptr(),
                      (const cb_bf16*)x.data_ptr(),
                      (const float*)gamma.data_ptr(),
@@ -452,7 +635,7 @@ torch::Tensor bn_bwd_dx(
                      (const float*)dgamma.data_ptr(),
                      (const float*)dbeta.data_ptr(),
                      (float)(1.0 / n), (cb_bf16*)dx.data_ptr(),
-                     total, C, training ? 1 : 0);
+                     M, C, training ? 1 : 0);
   CB_OK("bn_bwd_dx");
   return dx;
 }
@@ -463,13 +646,41 @@ torch::Tensor bnrelu_apply(
   TORCH_CHECK(x.scalar_type() == at::kBFloat16);
   const int C = x.size(1);
   auto xn = at::empty_like(x);
-  const long total = x.numel();
+  const long M = x.numel() / C;
+  const int pch = (int)std::min((M + 255) / 256, (long)128);
   hipLaunchKernelGGL(bnrelu_apply_kernel,
-                     dim3((total + 255) / 256), dim3(256), 0, cb_stream(),
+                     dim3(pch, (C + 7) / 8), dim3(256), 0, cb_stream(),
                      (const cb_bf16*)x.data_ptr(),
                      (const float*)scale.data_ptr(),
                      (const float*)shift.data_ptr(),
-                     (cb_bf16*)xn.data_ptr(), total, C);
+                     (cb_bf16*)xn.data_ptr(), M, C);
   CB_OK("bnrelu_apply");
   return xn;
+}
+
+std::vector<torch::Tensor> relu_mask_reduce(
+    torch::Tensor dxn, torch::Tensor x, torch::Tensor scale,
+    torch::Tensor shift, torch::Tensor mean, torch::Tensor invstd) {
+  TORCH_CHECK(dxn.is_contiguous(at::MemoryFormat::ChannelsLast));
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast));
+  const int C = x.size(1);
+  auto g = at::empty_like(x);
+  auto gb = torch::empty({2, (long)C}, scale.options());
+  cb_zero(gb);
+  auto dgamma = gb.select(0, 0);
+  auto dbeta = gb.select(0, 1);
+  const long M = x.numel() / C;
+  const int pch = (int)std::min((M + 255) / 256, (long)128);
+  hipLaunchKernelGGL(relu_mask_reduce_kernel,
+                     dim3(pch, (C + 7) / 8), dim3(256), 0, cb_stream(),
+                     (const cb_bf16*)dxn.data_ptr(),
+                     (const cb_bf16*)x.data_ptr(),
+                     (const float*)scale.data_ptr(),
+                     (const float*)shift.data_ptr(),
+                     (const float*)mean.data_ptr(),
+                     (const float*)invstd.data_ptr(),
+                     (cb_bf16*)g.data_ptr(), (float*)dgamma.data_ptr(),
+                     (float*)dbeta.data_ptr(), M, C);
+  CB_OK("relu_mask_reduce");
+  return {g, dgamma, dbeta};
 }

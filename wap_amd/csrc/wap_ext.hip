@@ -876,6 +876,9 @@ std::vector<torch::Tensor> bnconv_dgrad(
     torch::Tensor invstd, int64_t R, int64_t S);
 torch::Tensor bnrelu_apply(
     torch::Tensor x, torch::Tensor scale, torch::Tensor shift);
+std::vector<torch::Tensor> relu_mask_reduce(
+    torch::Tensor dxn, torch::Tensor x, torch::Tensor scale,
+    torch::Tensor shift, torch::Tensor mean, torch::Tensor invstd);
 torch::Tensor bn_bwd_dx(
     torch::Tensor g, torch::Tensor x, torch::Tensor gamma,
     torch::Tensor invstd, torch::Tensor mean, torch::Tensor dgamma,
@@ -888,6 +891,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_finalize", &bn_finalize);
   m.def("bnconv_dgrad", &bnconv_dgrad);
   m.def("bnrelu_apply", &bnrelu_apply);
+  m.def("relu_mask_reduce", &relu_mask_reduce);
   m.def("bn_bwd_dx", &bn_bwd_dx);
   m.def("bnrelu_conv_fwd", &bnrelu_conv_fwd);
   m.def("bnrelu_conv_fwd_fp8", &bnrelu_conv_fwd_fp8);

@@ -89,7 +89,14 @@ class _Transition(nn.Module):
         self.conv = nn.Conv2d(in_ch, out_ch, 1, bias=False)
 
     def forward(self, x: Tensor) -> Tensor:
-        return F.avg_pool2d(self.conv(F.relu(self.bn(x))), 2, ceil_mode=True)
+        from ..ops import fused_conv
+        if fused_conv.can_fuse_transition(x, self.conv.weight):
+            # hand stats/mask kernels + rocBLAS GEMMs (K > 96; aten's
+            # channels_last wrw here picked a ~5 ms CK kernel)
+            out = fused_conv.bnrelu_conv1x1_big(x, self.conv.weight, self.bn)
+        else:
+            out = self.conv(F.relu(self.bn(x)))
+        return F.avg_pool2d(out, 2, ceil_mode=True)
 
 
 class DenseNetEncoder(nn.Module):

@@ -145,3 +145,85 @@ def bnrelu_conv_fp8(x_q: Tensor, sx: Tensor, w_q: Tensor, sw: Tensor,
     return ext.require().bnrelu_conv_fwd_fp8(
         x_q, w_q, (scale.float() * sx).contiguous(),
         shift.float().contiguous(), float(sw), relu)
+
+
+class _BNReluLinear(torch.autograd.Function):
+    """Transition path: relu(bn(x)) -> 1x1 conv with K > 96, as hand
+    stats/mask kernels + rocBLAS GEMMs.
+
+    aten's channels_last wrw for these shapes picked a CK kernel measured
+    at ~5 ms/call (the single biggest kernel in the fused-step profile);
+    the GEMM route runs the same contraction in ~0.2 ms and the MIOpen BN
+    + relu + cast kernels disappear into ch_stats/bnrelu_apply.
+    """
+
+    @staticmethod
+    def forward(ctx, x: Tensor, weight: Tensor, gamma: Tensor, beta: Tensor,
+                running_mean: Tensor, running_var: Tensor, eps: float,
+                momentum: float, training: bool):
+        from . import ext
+        lib = ext.require()
+        B, C, H, W = x.shape
+        K = weight.shape[0]
+        if training:
+            sums, sumsq = lib.ch_stats(x)
+            n = x.numel() / C
+        else:
+            sums = sumsq = torch.empty(0)
+            n = 1.0
+        scale, shift, mean, invstd = lib.bn_finalize(
+            sums, sumsq, gamma.float(), beta.float(), running_mean,
+            running_var, float(n), eps, momentum if training else 0.0,
+            training)
+        xn = lib.bnrelu_apply(x, scale, shift)
+        M = B * H * W
+        xn2 = xn.permute(0, 2, 3, 1).reshape(M, C)      # free view (cl)
+        w2 = weight.squeeze(-1).squeeze(-1)             # [K, C]
+        out2 = xn2 @ w2.t()                             # rocBLAS [M, K]
+        out = out2.view(B, H, W, K).permute(0, 3, 1, 2)  # cl [B,K,H,W]
+        ctx.save_for_backward(x, w2, xn2, gamma, scale, shift, mean, invstd)
+        ctx.training = training
+        ctx.n = n
+        return out
+
+    @staticmethod
+    def backward(ctx, grad_out: Tensor):
+        from . import ext
+        lib = ext.require()
+        x, w2, xn2, gamma, scale, shift, mean, invstd = ctx.saved_tensors
+        B, C, H, W = x.shape
+        K = w2.shape[0]
+        M = B * H * W
+        grad_out = grad_out.contiguous(memory_format=torch.channels_last)
+        g2 = grad_out.permute(0, 2, 3, 1).reshape(M, K)
+        dxn2 = g2 @ w2                                  # [M, C]
+        dxn = dxn2.view(B, H, W, C).permute(0, 3, 1, 2)
+        g, dgamma, dbeta = lib.relu_mask_reduce(
+            dxn, x, scale, shift, mean, invstd)
+        dw = (g2.t() @ xn2).view(K, C, 1, 1)            # [K, C]
+        dx = lib.bn_bwd_dx(g, x, gamma.float(), invstd, mean, dgamma,
+                           dbeta, ctx.n, ctx.training)
+        return (dx, dw.to(w2.dtype), dgamma.to(gamma.dtype),
+                dbeta.to(gamma.dtype), None, None, None, None, None)
+
+
+def can_fuse_transition(x: Tensor, weight: Tensor) -> bool:
+    import os
+    if os.environ.get("WAP_FUSED_CONV", "auto") == "0":
+        return False
+    return (
+        use_hip(x)
+        and x.dtype == torch.bfloat16
+        and x.is_contiguous(memory_format=torch.channels_last)
+        and x.shape[1] <= 768
+        and weight.shape[2] == 1 and weight.shape[3] == 1
+    )
+
+
+def bnrelu_conv1x1_big(x: Tensor, weight: Tensor,
+                       bn: torch.nn.BatchNorm2d) -> Tensor:
+    """relu(bn(x)) -> 1x1 conv, any K (transition layers)."""
+    w = weight.to(torch.bfloat16).contiguous(memory_format=torch.channels_last)
+    return _BNReluLinear.apply(x, w, bn.weight, bn.bias, bn.running_mean,
+                               bn.running_var, bn.eps, bn.momentum,
+                               bn.training)

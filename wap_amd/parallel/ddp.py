@@ -115,6 +115,24 @@ class BucketedDDP:
                 b.buffer, op=dist.ReduceOp.SUM, group=self.group, async_op=True
             )
 
+    def clip_grad_norm_(self, max_norm: float) -> Tensor:
+        """Global-norm grad clip over the FLAT bucket buffers: a handful of
+        foreach kernels instead of per-parameter norm/mul launches (~400
+        params -> ~170 reduce + mul kernels per step with torch's clip).
+        Every trainable param's .grad is a view into a bucket, so scaling
+        the buffers scales the grads.  Matches
+        torch.nn.utils.clip_grad_norm_ semantics (scale by
+        max_norm/(total+1e-6), capped at 1)."""
+        bufs = [b.buffer for b in self.buckets]
+        if not bufs:
+            return torch.nn.utils.clip_grad_norm_(
+                self.model.parameters(), max_norm)
+        norms = torch._foreach_norm(bufs)
+        total = torch.linalg.vector_norm(torch.stack(norms))
+        scale = (max_norm / (total + 1e-6)).clamp(max=1.0)
+        torch._foreach_mul_(bufs, scale)
+        return total
+
     def finish_gradient_sync(self) -> None:
         if self.world_size <= 1:
             return

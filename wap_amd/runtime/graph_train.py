@@ -87,7 +87,7 @@ class GraphedTrainStep:
     def _clip_opt(self):
         import os
         if self.clip > 0 and os.environ.get("WAP_NO_CLIP", "0") != "1":
-            torch.nn.utils.clip_grad_norm_(self.model.parameters(), self.clip)
+            self.ddp.clip_grad_norm_(self.clip)
         self.opt.step()
 
     def _allreduce(self):
@@ -248,7 +248,7 @@ class MultiShapeGraphedTrainer:
 
     def _clip_opt_ops(self):
         if self.clip > 0:
-            torch.nn.utils.clip_grad_norm_(self.model.parameters(), self.clip)
+            self.ddp.clip_grad_norm_(self.clip)
         self.opt.step()
 
     def _opt_step(self):
@@ -260,9 +260,7 @@ class MultiShapeGraphedTrainer:
         safe_stream_sync()
         self.g_opt = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.g_opt, capture_error_mode=_capture_mode()):
-            if self.clip > 0:
-                torch.nn.utils.clip_grad_norm_(self.model.parameters(), self.clip)
-            self.opt.step()
+            self._clip_opt_ops()
         self.g_opt.replay()
 
     def _eager_fwd_bwd(self, batch):
