@@ -281,27 +281,21 @@ void bnconv_dgrad_kernel(
     float* __restrict__ dgamma, float* __restrict__ dbeta,
     int Bn, int H, int W, int C, int K, int Kp, int R, int S) {
   // Whole reduction axis (all taps x all K-chunks, R*S*Kp <= 288) staged
-  // ONCE: two barriers for the entire block (a per-(tap,chunk) staging
-  // loop measured ~70us/call = 3x MIOpen igemm_bwd).  Each thread owns a
-  // fixed pixel row of the tile, so the p -> (b,y,x) division happens
-  // once, not once per staged piece.
+  // ONCE, then the block loops over ALL channel tiles reusing the staged
+  // g_out — grid is (M-tiles) only.  A (M-tile x C-tile) grid re-read
+  // g_out C/64 times (138 MB per big 1x1 layer) and lost to MIOpen's
+  // igemm_bwd.  Each thread owns a fixed pixel row of the tile, so the
+  // p -> (b,y,x) division happens once.
   __shared__ cb_bf16 lG[DG_BM][296];       // up to 9 taps x 32 K, +8 pad
   const int HW = H * W;
   const long M = (long)Bn * HW;
   const long m0 = (long)blockIdx.x * DG_BM;
-  const int c0 = blockIdx.y * DG_BN;
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
   const int wv = tid >> 6;
   const int PR = R / 2, PS = S / 2;
   const int Cp = ((C + DG_BN - 1) / DG_BN) * DG_BN;
   const int TK = R * S * Kp;               // staged row length (<= 288)
-
-  cb_f32x4 acc[2][4];
-#pragma unroll
-  for (int i = 0; i < 2; ++i)
-#pragma unroll
-    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   {
     // threads t and t+128 share pixel row t%128, staging half its pieces
@@ -347,62 +341,65 @@ void bnconv_dgrad_kernel(
     }
   }
   __syncthreads();
-  for (int tap = 0; tap < R * S; ++tap) {
-    const long wtap = ((long)tap * Cp) * Kp;
-    for (int kc = 0; kc < Kp; kc += 32) {
+
+  for (int c0 = 0; c0 < C; c0 += DG_BN) {
+    cb_f32x4 acc[2][4];
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+    for (int tap = 0; tap < R * S; ++tap) {
+      const long wtap = ((long)tap * Cp) * Kp;
+      for (int kc = 0; kc < Kp; kc += 32) {
+#pragma unroll
+        for (int mf = 0; mf < 2; ++mf) {
+          const cb_bf16x8 afrag = *(const cb_bf16x8*)
+              &lG[wv * 32 + mf * 16 + (lane & 15)][tap * Kp + kc +
+                                                   (lane >> 4) * 8];
+#pragma unroll
+          for (int nf = 0; nf < 4; ++nf) {
+            const cb_bf16x8 bfrag = *(const cb_bf16x8*)
+                &wT[wtap + (long)(c0 + nf * 16 + (lane & 15)) * Kp + kc +
+                    (lane >> 4) * 8];
+            acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afrag, bfrag, acc[mf][nf], 0, 0, 0);
+          }
+        }
+      }
+    }
+    // epilogue for this channel tile: relu mask from x, store g, reduce
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      const int c = c0 + nf * 16 + (lane & 15);
+      const bool cv = c < C;
+      const float sc = cv ? scale[c] : 0.f;
+      const float sh = cv ? shift[c] : 0.f;
+      const float mn = cv ? mean[c] : 0.f;
+      const float is = cv ? invstd[c] : 0.f;
+      float db = 0.f, dg = 0.f;
 #pragma unroll
       for (int mf = 0; mf < 2; ++mf) {
-        const cb_bf16x8 afrag = *(const cb_bf16x8*)
-            &lG[wv * 32 + mf * 16 + (lane & 15)][tap * Kp + kc +
-                                                 (lane >> 4) * 8];
 #pragma unroll
-        for (int nf = 0; nf < 4; ++nf) {
-          const cb_bf16x8 bfrag = *(const cb_bf16x8*)
-              &wT[wtap + (long)(c0 + nf * 16 + (lane & 15)) * Kp + kc +
-                  (lane >> 4) * 8];
-          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              afrag, bfrag, acc[mf][nf], 0, 0, 0);
+        for (int reg = 0; reg < 4; ++reg) {
+          const long p = m0 + wv * 32 + mf * 16 + (lane >> 4) * 4 + reg;
+          if (cv && p < M) {
+            const float xv = cb_b2f(x[p * C + c]);
+            const float gg =
+                (xv * sc + sh) > 0.f ? acc[mf][nf][reg] : 0.f;
+            g[p * C + c] = cb_f2b(gg);
+            db += gg;
+            dg += gg * (xv - mn) * is;
+          }
         }
       }
-    }
-  }
-  __syncthreads();
-
-  // epilogue: relu mask from x, store g, reduce dbeta/dgamma
-  // (direct per-accumulator form; an LDS-repacked variant with 16-byte
-  // row IO measured SLOWER — 9.9 vs 5.9 ms/step — the extra barrier +
-  // register pressure outweigh the coalescing win at these tile sizes)
-#pragma unroll
-  for (int nf = 0; nf < 4; ++nf) {
-    const int c = c0 + nf * 16 + (lane & 15);
-    const bool cv = c < C;
-    const float sc = cv ? scale[c] : 0.f;
-    const float sh = cv ? shift[c] : 0.f;
-    const float mn = cv ? mean[c] : 0.f;
-    const float is = cv ? invstd[c] : 0.f;
-    float db = 0.f, dg = 0.f;
-#pragma unroll
-    for (int mf = 0; mf < 2; ++mf) {
-#pragma unroll
-      for (int reg = 0; reg < 4; ++reg) {
-        const long p = m0 + wv * 32 + mf * 16 + (lane >> 4) * 4 + reg;
-        if (cv && p < M) {
-          const float xv = cb_b2f(x[p * C + c]);
-          const float gg =
-              (xv * sc + sh) > 0.f ? acc[mf][nf][reg] : 0.f;
-          g[p * C + c] = cb_f2b(gg);
-          db += gg;
-          dg += gg * (xv - mn) * is;
-        }
+      db += __shfl_xor(db, 16, WAVE);
+      db += __shfl_xor(db, 32, WAVE);
+      dg += __shfl_xor(dg, 16, WAVE);
+      dg += __shfl_xor(dg, 32, WAVE);
+      if ((lane >> 4) == 0 && cv) {
+        atomicAdd(dbeta + c, db);
+        atomicAdd(dgamma + c, dg);
       }
-    }
-    db += __shfl_xor(db, 16, WAVE);
-    db += __shfl_xor(db, 32, WAVE);
-    dg += __shfl_xor(dg, 16, WAVE);
-    dg += __shfl_xor(dg, 32, WAVE);
-    if ((lane >> 4) == 0 && cv) {
-      atomicAdd(dbeta + c, db);
-      atomicAdd(dgamma + c, dg);
     }
   }
 }
@@ -547,7 +544,7 @@ std::vector<torch::Tensor> bnconv_dgrad(
   auto dgamma = gb.select(0, 0);
   auto dbeta = gb.select(0, 1);
   const long M = (long)Bn * H * W;
-  const dim3 grid((M + DG_BM - 1) / DG_BM, (C + DG_BN - 1) / DG_BN);
+  const dim3 grid((M + DG_BM - 1) / DG_BM);
   hipLaunchKernelGGL(bnconv_dgrad_kernel, grid, dim3(256), 0, cb_stream(),
                      (const cb_bf16*)gout.data_ptr(),
                      (const cb_bf16*)wT.data_ptr(),

@@ -100,7 +100,7 @@ def can_fuse(x: Tensor, weight: Tensor) -> bool:
     # the backward is now hand-written too.  WAP_FUSED_CONV=0 disables,
     # =1x1 restricts to the 1x1 convs (A/B debugging).
     mode = os.environ.get("WAP_FUSED_CONV", "auto")
-    if mode == "0":
+    if mode in ("0", "trans"):   # "trans": transitions only, for A/B
         return False
     if mode == "1x1" and weight.shape[2] != 1:
         return False
