@@ -279,13 +279,14 @@ void bnconv_dgrad_kernel(
     const float* __restrict__ shift, const float* __restrict__ mean,
     const float* __restrict__ invstd, cb_bf16* __restrict__ g,
     float* __restrict__ dgamma, float* __restrict__ dbeta,
-    int Bn, int H, int W, int C, int K, int Kp, int R, int S) {
+    int Bn, int H, int W, int C, int K, int Kp, int R, int S, int cchunk) {
   // Whole reduction axis (all taps x all K-chunks, R*S*Kp <= 288) staged
-  // ONCE, then the block loops over ALL channel tiles reusing the staged
-  // g_out — grid is (M-tiles) only.  A (M-tile x C-tile) grid re-read
-  // g_out C/64 times (138 MB per big 1x1 layer) and lost to MIOpen's
-  // igemm_bwd.  Each thread owns a fixed pixel row of the tile, so the
-  // p -> (b,y,x) division happens once.
+  // ONCE, then the block loops over `cchunk` channel tiles reusing the
+  // staged g_out.  cchunk is picked on the host so the grid stays >= 512
+  // blocks: a pure (M x C) grid re-read g_out C/64 times (138 MB on the
+  // big 1x1 layers), a pure M grid starved the chip on the small-spatial
+  // blocks (32 blocks).  Each thread owns a fixed pixel row of the tile,
+  // so the p -> (b,y,x) division happens once.
   __shared__ cb_bf16 lG[DG_BM][296];       // up to 9 taps x 32 K, +8 pad
   const int HW = H * W;
   const long M = (long)Bn * HW;
@@ -342,7 +343,9 @@ void bnconv_dgrad_kernel(
   }
   __syncthreads();
 
-  for (int c0 = 0; c0 < C; c0 += DG_BN) {
+  const int cstart = blockIdx.y * cchunk * DG_BN;
+  const int cend = min(C, cstart + cchunk * DG_BN);
+  for (int c0 = cstart; c0 < cend; c0 += DG_BN) {
     cb_f32x4 acc[2][4];
 #pragma unroll
     for (int i = 0; i < 2; ++i)
@@ -544,7 +547,11 @@ std::vector<torch::Tensor> bnconv_dgrad(
   auto dgamma = gb.select(0, 0);
   auto dbeta = gb.select(0, 1);
   const long M = (long)Bn * H * W;
-  const dim3 grid((M + DG_BM - 1) / DG_BM);
+  const long mtiles = (M + DG_BM - 1) / DG_BM;
+  const int ctiles = (C + DG_BN - 1) / DG_BN;
+  int cchunk = (int)std::max((long)1, mtiles * ctiles / 512);
+  cchunk = std::min(cchunk, ctiles);
+  const dim3 grid(mtiles, (ctiles + cchunk - 1) / cchunk);
   hipLaunchKernelGGL(bnconv_dgrad_kernel, grid, dim3(256), 0, cb_stream(),
                      (const cb_bf16*)gout.data_ptr(),
                      (const cb_bf16*)wT.data_ptr(),
@@ -555,7 +562,7 @@ std::vector<torch::Tensor> bnconv_dgrad(
                      (const float*)invstd.data_ptr(),
                      (cb_bf16*)g.data_ptr(), (float*)dgamma.data_ptr(),
                      (float*)dbeta.data_ptr(),
-                     Bn, H, W, C, K, Kp, (int)R, (int)S);
+                     Bn, H, W, C, K, Kp, (int)R, (int)S, cchunk);
   CB_OK("bnconv_dgrad");
   return {g, dgamma, dbeta};
 }
