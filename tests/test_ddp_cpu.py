@@ -113,3 +113,57 @@ def test_frozen_param_buckets_still_sync(tmp_path):
     g1 = torch.load(tmp_path / "fg_1.pt", weights_only=False)
     for n in g0:
         assert torch.equal(g0[n], g1[n]), n
+
+
+def _trainer_worker(rank, world, port, out_dir):
+    """World-4 collective-alignment test of MultiShapeGraphedTrainer's
+    step() control flow (graphs_enabled=False: eager stand-ins execute,
+    the cache/capture/fallback branching and the collective placement are
+    the real code).  Ranks deliberately DIVERGE in caching behavior —
+    rank 0 never caches (max_graphs=0, always the eager-fallback branch),
+    others cache per shape — while batches cycle through several shapes.
+    Params must stay identical across ranks after every step."""
+    from wap_amd import ops
+    ops.set_backend("eager")
+    dist.init_process_group(
+        "gloo", init_method=f"tcp://127.0.0.1:{port}", rank=rank,
+        world_size=world
+    )
+    from wap_amd.parallel import BucketedDDP
+    from wap_amd.runtime.graph_train import MultiShapeGraphedTrainer
+
+    m = _mk_model()
+    ddp = BucketedDDP(m, bucket_bytes=1 << 18)
+    opt = torch.optim.Adadelta(m.parameters(), lr=1.0)
+    trainer = MultiShapeGraphedTrainer(
+        m, ddp, opt, clip=1.0, autocast_bf16=False,
+        max_graphs=0 if rank == 0 else 8, graphs_enabled=False,
+    )
+    shapes = [(32, 32, 4), (32, 48, 5), (48, 32, 4), (32, 32, 6)]
+    g = torch.Generator().manual_seed(500 + rank)
+    for step in range(6):
+        h, w, t = shapes[step % len(shapes)]
+        batch = {
+            "x": torch.rand(2, 1, h, w, generator=g),
+            "x_mask": torch.ones(2, h, w),
+            "y": torch.randint(1, 10, (2, t), generator=g),
+            "y_mask": torch.ones(2, t),
+        }
+        trainer.step(batch)
+        # params identical across ranks after every step
+        for n, p in m.named_parameters():
+            ref = p.data.clone()
+            dist.broadcast(ref, src=0)
+            assert torch.allclose(p.data, ref, atol=1e-6), (step, n, rank)
+    if rank == 0:
+        assert trainer.n_eager_fallback > 0      # rank 0 took the fallback
+    else:
+        assert len(trainer.cache) > 0            # others took the cache
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def test_multishape_trainer_world4_alignment(tmp_path):
+    port = 29737
+    mp.spawn(_trainer_worker, args=(4, port, str(tmp_path)), nprocs=4,
+             join=True)

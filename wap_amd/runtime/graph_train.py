@@ -32,8 +32,10 @@ def safe_stream_sync() -> None:
     hipDeviceSynchronize: on ROCm 7.x those corrupt the next hipGraph
     replay's backward (measured, deterministic on most boxes).  A D2H
     scalar read (hipMemcpyWithStream under .item()) drains the stream and
-    is measured safe."""
+    is measured safe.  No-op on CPU (nothing to drain)."""
     global _SYNC_SCRATCH
+    if not torch.cuda.is_available():
+        return
     if _SYNC_SCRATCH is None or not _SYNC_SCRATCH.is_cuda:
         _SYNC_SCRATCH = torch.zeros(1, device="cuda")
     _SYNC_SCRATCH.item()
@@ -203,6 +205,24 @@ class _ShapeGraph:
         return out["loss"]
 
 
+class _EagerShapeStep:
+    """Capture-free stand-in for _ShapeGraph (same interface): executes
+    fwd+bwd eagerly on replay().  Used when graphs are disabled so the
+    trainer's step() control flow — shape cache, capture branch, eager
+    fallback, collective placement — runs (and is testable) on CPU."""
+
+    def __init__(self, trainer: "MultiShapeGraphedTrainer",
+                 batch: Dict[str, torch.Tensor]):
+        self.t = trainer
+        self.buf = {k: batch[k].clone() for k in ("x", "x_mask", "y", "y_mask")}
+        self.loss: Optional[torch.Tensor] = None
+        self.has_opt = False
+        self.graph = self
+
+    def replay(self) -> None:
+        self.loss = self.t._eager_fwd_bwd(self.buf)
+
+
 class MultiShapeGraphedTrainer:
     """Graphed training over VARIABLE bucket shapes (train.py).
 
@@ -220,7 +240,8 @@ class MultiShapeGraphedTrainer:
     """
 
     def __init__(self, model, ddp, optimizer, clip: float = 100.0,
-                 autocast_bf16: bool = True, max_graphs: int = 24):
+                 autocast_bf16: bool = True, max_graphs: int = 24,
+                 graphs_enabled: Optional[bool] = None):
         self.model = model
         self.ddp = ddp
         self.opt = optimizer
@@ -231,9 +252,17 @@ class MultiShapeGraphedTrainer:
         self._failed: set = set()
         self.g_opt: Optional[torch.cuda.CUDAGraph] = None
         self.n_eager_fallback = 0
+        # graphs_enabled=False runs the IDENTICAL step() control flow
+        # (shape cache, capture branch, eager fallback, collective
+        # placement) with eager execution standing in for capture/replay —
+        # this is what the CPU world-4 collective-alignment test exercises
+        # (tests/test_ddp_cpu.py), so the distributed step logic is covered
+        # without a GPU.
+        self.graphs_enabled = (torch.cuda.is_available()
+                               if graphs_enabled is None else graphs_enabled)
         ddp.sync_enabled = False
         for g in self.opt.param_groups:
-            g["capturable"] = True
+            g["capturable"] = self.graphs_enabled
 
     def _allreduce(self):
         if self.ddp.world_size > 1:
@@ -252,6 +281,9 @@ class MultiShapeGraphedTrainer:
         self.opt.step()
 
     def _opt_step(self):
+        if not self.graphs_enabled:
+            self._clip_opt_ops()
+            return
         if self.g_opt is not None:
             self.g_opt.replay()
             return
@@ -282,11 +314,14 @@ class MultiShapeGraphedTrainer:
             # stream, and default-stream-bound AccumulateGrad breaks later
             # graph captures (segfault in capture_end; torch warns about
             # exactly this).
-            s = torch.cuda.Stream()
-            s.wait_stream(torch.cuda.current_stream())
-            with torch.cuda.stream(s):
+            if self.graphs_enabled:
+                s = torch.cuda.Stream()
+                s.wait_stream(torch.cuda.current_stream())
+                with torch.cuda.stream(s):
+                    loss = self._eager_fwd_bwd(batch)
+                torch.cuda.current_stream().wait_stream(s)
+            else:
                 loss = self._eager_fwd_bwd(batch)
-            torch.cuda.current_stream().wait_stream(s)
             self._allreduce()
             self._clip_opt_ops()
             return loss
@@ -304,7 +339,8 @@ class MultiShapeGraphedTrainer:
             # the Adadelta state dict is lazily created at the first eager
             # step, so the opt graph is only captured from step 2 on.
             try:
-                sg = _ShapeGraph(self, batch)
+                sg = (_ShapeGraph(self, batch) if self.graphs_enabled
+                      else _EagerShapeStep(self, batch))
                 self.cache[key] = sg
                 sg.graph.replay()  # capture records only; compute this batch
                 if sg.has_opt:
