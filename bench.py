@@ -185,6 +185,24 @@ def main() -> None:
         extra["decode_tok_s"] = tok / td
         extra["decode_beam"] = args.beam
         extra["decode_batch"] = db
+        # contract-length decode (reference maxlen=200):
+        # a random-init model rarely emits eos, so this measures the
+        # full-length beam loop, not early termination
+        with autocast():
+            beam_decode(model, x[:db], x_mask[:db], beam=args.beam,
+                        maxlen=200)  # warm the maxlen-200 graph shapes
+        if on_gpu:
+            safe_stream_sync()
+        tok2 = 0
+        td0 = time.perf_counter()
+        for _ in range(decode_rounds):
+            with autocast():
+                hyps = beam_decode(model, x[:db], x_mask[:db],
+                                   beam=args.beam, maxlen=200)
+            tok2 += sum(max(len(h), 1) for h in hyps)
+        if on_gpu:
+            safe_stream_sync()
+        extra["decode_tok_s_maxlen200"] = tok2 / (time.perf_counter() - td0)
 
     if rank == 0:
         print(json.dumps({

@@ -78,17 +78,21 @@ def load_split(cfg: WAPConfig, vocab: Vocab, split: str, seed: int):
 def validate(model, buckets, vocab, cfg, device, autocast_ctx) -> float:
     """Greedy-decode the validation buckets -> WER (rank 0 only)."""
     pairs = []
-    pad_to = ((cfg.graph_pad_hw, cfg.graph_pad_hw, cfg.graph_pad_t, 1)
-              if torch.device(device).type == "cuda" else None)
+    on_gpu = torch.device(device).type == "cuda"
+    # pad validation buckets onto the SAME shape grid training uses,
+    # including the batch dim, so the hipGraph-captured decode step is
+    # reused across buckets (a handful of captures per run instead of
+    # per-shape eager loops)
+    pad_to = ((cfg.graph_pad_hw, cfg.graph_pad_hw, cfg.graph_pad_t,
+               cfg.batch_size) if on_gpu else None)
     loader = BucketLoader(buckets, device=device, pad_to=pad_to)
     with torch.no_grad():
         for batch in loader:
             with autocast_ctx():
-                # use_graph=False: validation buckets have many distinct
-                # shapes, per-shape capture would dominate
                 hyps = greedy_decode(
                     model, batch["x"], batch["x_mask"],
-                    maxlen=cfg.decode_maxlen, fused=True, use_graph=False,
+                    maxlen=cfg.decode_maxlen, fused=True,
+                    use_graph=on_gpu,
                 )
             refs = batch["y"]
             masks = batch["y_mask"]
