@@ -62,32 +62,41 @@ class _BNReluConv(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, grad_out: Tensor):
+        import os
         from . import ext
         lib = ext.require()
         x, weight, gamma, scale, shift, mean, invstd = ctx.saved_tensors
         grad_out = grad_out.contiguous(memory_format=torch.channels_last)
         K, C, R, S = weight.shape
-        Kp = ((K + 31) // 32) * 32
-        Cp = ((C + 63) // 64) * 64
-        # wT [R*S, Cp, Kp] from the channels_last weight (physical KRSC):
-        # one pad kernel (materializes the permuted view zero-padded)
-        wphys = weight.permute(0, 2, 3, 1).reshape(K, R * S, C)
-        wT = torch.nn.functional.pad(wphys.permute(1, 2, 0),
-                                     (0, Kp - K, 0, Cp - C)).contiguous()
-        g, dgamma, dbeta = lib.bnconv_dgrad(
-            grad_out, wT, x, scale, shift, mean, invstd, R, S)
-        # wrw: materialize xn once (one elementwise kernel), then it is a
-        # plain GEMM — rocBLAS for 1x1, MIOpen igemm for 3x3.  The
-        # hand-written LDS-transposed wrw kernel measured 1.4 ms/call vs
-        # ~20 us here (PERF.md): reduction over the fast (channel-adjacent)
-        # pixel dim wants a library GEMM, not a per-tile transpose.
-        xn = lib.bnrelu_apply(x, scale, shift)
         pad = R // 2
-        # MIOpen igemm wrw (~19us at these shapes; rocBLAS on the same
-        # tall-skinny [K,M]x[M,C] TN GEMM measured 166us — no split-K)
-        _, dwl, _ = torch.ops.aten.convolution_backward(
-            grad_out, xn, weight, None, [1, 1], [pad, pad], [1, 1],
-            False, [0, 0], 1, [False, True, False])
+        # xn materialized once (one elementwise kernel) for the library
+        # conv grads.  A hand LDS-transposed wrw kernel measured 1.4
+        # ms/call vs ~20 us for MIOpen igemm (PERF.md): the reduction over
+        # the channel-adjacent pixel dim wants a library GEMM.
+        xn = lib.bnrelu_apply(x, scale, shift)
+        if os.environ.get("WAP_FUSED_CONV", "0") == "hybrid":
+            # hybrid: MIOpen igemm for BOTH conv grads (its dgrad still
+            # beats bnconv_dgrad), our kernels for the relu mask + BN
+            # reductions + BN backward
+            dxn, dwl, _ = torch.ops.aten.convolution_backward(
+                grad_out, xn, weight, None, [1, 1], [pad, pad], [1, 1],
+                False, [0, 0], 1, [True, True, False])
+            g, dgamma, dbeta = lib.relu_mask_reduce(
+                dxn.contiguous(memory_format=torch.channels_last), x,
+                scale, shift, mean, invstd)
+        else:
+            Kp = ((K + 31) // 32) * 32
+            Cp = ((C + 63) // 64) * 64
+            # wT [R*S, Cp, Kp] from the cl weight (physical KRSC): one pad
+            # kernel materializes the permuted view zero-padded
+            wphys = weight.permute(0, 2, 3, 1).reshape(K, R * S, C)
+            wT = torch.nn.functional.pad(wphys.permute(1, 2, 0),
+                                         (0, Kp - K, 0, Cp - C)).contiguous()
+            g, dgamma, dbeta = lib.bnconv_dgrad(
+                grad_out, wT, x, scale, shift, mean, invstd, R, S)
+            _, dwl, _ = torch.ops.aten.convolution_backward(
+                grad_out, xn, weight, None, [1, 1], [pad, pad], [1, 1],
+                False, [0, 0], 1, [False, True, False])
         dx = lib.bn_bwd_dx(g, x, gamma.float(), invstd, mean, dgamma, dbeta,
                            ctx.n, ctx.training)
         return (dx, dwl.to(weight.dtype), dgamma.to(gamma.dtype),
@@ -108,6 +117,7 @@ def can_fuse(x: Tensor, weight: Tensor) -> bool:
         return False
     if mode == "1x1" and weight.shape[2] != 1:
         return False
+    # "hybrid": fused forward + MIOpen conv backward (see backward())
     return (
         use_hip(x)
         and x.dtype == torch.bfloat16
