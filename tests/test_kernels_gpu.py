@@ -490,3 +490,25 @@ def test_bucket_clip_matches_torch(nbuf):
     assert torch.allclose(n1, n2, rtol=1e-5)
     for p1, p2 in zip(m1.parameters(), m2.parameters()):
         assert torch.allclose(p1.grad, p2.grad, rtol=1e-5, atol=1e-7)
+
+
+@pytest.mark.parametrize("B,C,H,W,K", [(4, 432, 16, 32, 96),
+                                       (2, 684, 8, 32, 96)])
+def test_mxfp8_conv_matches_fp8(B, C, H, W, K):
+    """The MX-scaled K=128 fp8 conv (unit block scales) must match the
+    non-scaled fp8 kernel — same math, faster instruction."""
+    from wap_amd.ops import ext
+    from wap_amd.ops.fused_conv import quantize_e4m3
+    m = ext.require()
+    torch.manual_seed(51)
+    x = (torch.randn(B, C, H, W, device="cuda") * 0.5)
+    w = (torch.randn(K, C, 1, 1, device="cuda") * 0.05)
+    scale = torch.rand(C, device="cuda") + 0.5
+    shift = torch.randn(C, device="cuda") * 0.3
+    x_q, sx = quantize_e4m3(x)
+    w_q, sw = quantize_e4m3(w)
+    ref = m.bnrelu_conv_fwd_fp8(x_q, w_q, (scale * sx).contiguous(),
+                                shift.contiguous(), float(sw), True)
+    mx = m.bnrelu_conv_fwd_mxfp8(x_q, w_q, (scale * sx).contiguous(),
+                                 shift.contiguous(), float(sw), True)
+    assert _rel_err(mx, ref) < 1e-2, _rel_err(mx, ref)

@@ -38,7 +38,8 @@ def timeit(fn, iters=50):
 def main():
     m = ext.require()
     dev = "cuda"
-    print(f"{'shape':>28} {'bf16 us':>9} {'fp8 us':>9} {'miopen us':>10} {'ratio':>6}")
+    print(f"{'shape':>28} {'bf16 us':>9} {'fp8 us':>9} {'mxfp8 us':>9} "
+          f"{'miopen us':>10} {'ratio':>6}")
     for (B, C, H, W, K, R) in SHAPES:
         x = (torch.randn(B, C, H, W, device=dev) * 0.5).to(torch.bfloat16) \
             .contiguous(memory_format=torch.channels_last)
@@ -65,9 +66,15 @@ def main():
         t_fp8 = timeit(lambda: m.bnrelu_conv_fwd_fp8(
             x_q, w_q, (scale * sx).contiguous(), shift.contiguous(),
             float(sw), True))
+        t_mx = float("nan")
+        if R == 1:
+            t_mx = timeit(lambda: m.bnrelu_conv_fwd_mxfp8(
+                x_q, w_q, (scale * sx).contiguous(), shift.contiguous(),
+                float(sw), True))
         flops = 2 * B * H * W * K * C * R * R
-        print(f"{(B,C,H,W,K,R)!s:>28} {t_mfma:9.1f} {t_fp8:9.1f} {t_ref:10.1f} "
-              f"{t_ref/t_mfma:6.2f}  ({flops/t_mfma/1e6:.0f} GF/s mfma)")
+        print(f"{(B,C,H,W,K,R)!s:>28} {t_mfma:9.1f} {t_fp8:9.1f} {t_mx:9.1f} "
+              f"{t_ref:10.1f} {t_ref/t_mfma:6.2f}  "
+              f"({flops/t_mfma/1e6:.0f} GF/s mfma)")
 
 
 if __name__ == "__main__":

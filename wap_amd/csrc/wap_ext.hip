@@ -783,6 +783,9 @@ extern "C" __global__ void bnrelu_conv_mfma_fp8_fwd(
 extern "C" __global__ void bnrelu_conv_mfma_fp8_fwd_bk96(
     const __hip_fp8_e4m3*, const __hip_fp8_e4m3*, const float*, const float*,
     float, __hip_bfloat16*, int, int, int, int, int, int, int, int);
+extern "C" __global__ void bnrelu_conv_mfma_mx_fwd(
+    const __hip_fp8_e4m3*, const __hip_fp8_e4m3*, const float*, const float*,
+    float, __hip_bfloat16*, int, int, int, int, int, int);
 
 torch::Tensor bnrelu_conv_fwd(
     torch::Tensor x, torch::Tensor w, torch::Tensor scale,
@@ -884,7 +887,34 @@ torch::Tensor bn_bwd_dx(
     torch::Tensor invstd, torch::Tensor mean, torch::Tensor dgamma,
     torch::Tensor dbeta, double n, bool training);
 
+torch::Tensor bnrelu_conv_fwd_mxfp8(
+    torch::Tensor x, torch::Tensor w, torch::Tensor scale,
+    torch::Tensor shift, double out_scale, bool relu_on) {
+  TORCH_CHECK(x.is_cuda() && w.is_cuda());
+  TORCH_CHECK(x.scalar_type() == at::kFloat8_e4m3fn &&
+              w.scalar_type() == at::kFloat8_e4m3fn);
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast) &&
+              w.is_contiguous(at::MemoryFormat::ChannelsLast));
+  const int B = x.size(0), C = x.size(1), H = x.size(2), W = x.size(3);
+  const int K = w.size(0);
+  TORCH_CHECK(w.size(1) == C && w.size(2) == 1 && w.size(3) == 1 && K <= 96,
+              "mxfp8 path: 1x1 conv, K <= 96");
+  auto out = at::empty({B, K, H, W}, x.options().dtype(at::kBFloat16),
+                       at::MemoryFormat::ChannelsLast);
+  const long M = (long)B * H * W;
+  hipLaunchKernelGGL(bnrelu_conv_mfma_mx_fwd, dim3((M + 127) / 128),
+                     dim3(256), 0, cur_stream(),
+                     reinterpret_cast<const __hip_fp8_e4m3*>(x.data_ptr()),
+                     reinterpret_cast<const __hip_fp8_e4m3*>(w.data_ptr()),
+                     cptr<float>(scale), cptr<float>(shift),
+                     (float)out_scale, ptr<__hip_bfloat16>(out),
+                     B, H, W, C, K, relu_on ? 1 : 0);
+  HIP_OK();
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("bnrelu_conv_fwd_mxfp8", &bnrelu_conv_fwd_mxfp8);
   m.def("fattn_score_fwd", &fattn_score_fwd);
   m.def("fattn_score_bwd", &fattn_score_bwd);
   m.def("ch_stats", &ch_stats);
