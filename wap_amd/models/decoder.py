@@ -160,7 +160,23 @@ class Parser(nn.Module):
         w_qh = torch.cat([att.W_a.weight, self.gru2.w_h], dim=0)  # [A+3n, n]
         b_qh = torch.cat([self.gru2.b_h.new_zeros(A), self.gru2.b_h])
         w_cov = att.merged_coverage_weight()
+        import os
         from ..ops import functional as opf
+        if os.environ.get("WAP_MANUAL_BPTT", "1") != "0":
+            # hand-written BPTT over the unroll (ops/unroll.py): same
+            # per-step kernels, but the backward batches every recurrent
+            # weight grad into one GEMM and keeps carried grads in
+            # preallocated buffers (the autograd tape's per-step grad
+            # GEMMs + accumulation adds dominate the dispatch count)
+            from ..ops.unroll import decoder_unroll
+            dt = a.dtype
+            s_all, c_all = decoder_unroll(
+                gx1_all.to(dt), s.to(dt), a_flat, ua, mask_flat,
+                w_qh.to(dt), b_qh.to(dt), self.gru2.w_x.to(dt),
+                self.gru2.b_x.to(dt), self.gru1.w_h.to(dt),
+                self.gru1.b_h.to(dt), w_cov, self.attention.U_f.bias,
+                self.attention.nu, hp, wp)
+            return self.readout(s_all, c_all, emb_all)
         w_pack = (opf.pack_coverage_weight(w_cov)
                   if opf.can_fuse_cov_score(coverage, w_cov, ua) else None)
         ss, cs = [], []
