@@ -512,3 +512,42 @@ def test_mxfp8_conv_matches_fp8(B, C, H, W, K):
     mx = m.bnrelu_conv_fwd_mxfp8(x_q, w_q, (scale * sx).contiguous(),
                                  shift.contiguous(), float(sw), True)
     assert _rel_err(mx, ref) < 1e-2, _rel_err(mx, ref)
+
+
+@pytest.mark.parametrize("B,C,H,W", [(4, 48, 32, 128), (2, 684, 8, 32),
+                                     (3, 96, 7, 13)])
+@pytest.mark.parametrize("training", [True, False])
+def test_fused_bnrelu_nchw_autograd(B, C, H, W, training):
+    """NCHW BN+ReLU elementwise fusion vs fp32 eager, fwd + all grads."""
+    from wap_amd.ops.fused_conv import _BNRelu
+    torch.manual_seed(61)
+    dev = "cuda"
+    x0 = torch.randn(B, C, H, W, device=dev)
+    gamma0 = torch.rand(C, device=dev) + 0.5
+    beta0 = torch.randn(C, device=dev) * 0.3
+    rm0 = torch.randn(C, device=dev) * 0.1
+    rv0 = torch.rand(C, device=dev) + 0.5
+
+    x = x0.clone().requires_grad_()
+    gamma = gamma0.clone().requires_grad_()
+    beta = beta0.clone().requires_grad_()
+    rm, rv = rm0.clone(), rv0.clone()
+    ref = torch.relu(torch.nn.functional.batch_norm(
+        x, rm, rv, gamma, beta, training=training, momentum=0.1, eps=1e-5))
+    torch.manual_seed(63)
+    gout = torch.randn_like(ref)
+    ref.backward(gout)
+
+    xh = x0.clone().to(torch.bfloat16).requires_grad_()
+    gammah = gamma0.clone().requires_grad_()
+    betah = beta0.clone().requires_grad_()
+    rmh, rvh = rm0.clone(), rv0.clone()
+    out = _BNRelu.apply(xh, gammah, betah, rmh, rvh, 1e-5, 0.1, training)
+    out.backward(gout.to(torch.bfloat16))
+
+    assert _rel_err(out, ref) < 3e-2
+    assert _rel_err(rmh, rm) < 2e-2 and _rel_err(rvh, rv) < 2e-2
+    for name, h, r in [("dx", xh.grad, x.grad),
+                       ("dgamma", gammah.grad, gamma.grad),
+                       ("dbeta", betah.grad, beta.grad)]:
+        assert _rel_err(h, r) < 6e-2, (name, _rel_err(h, r))

@@ -638,3 +638,254 @@ std::vector<torch::Tensor> relu_mask_reduce(
   CB_OK("relu_mask_reduce");
   return {g, dgamma, dbeta};
 }
+
+// ===================================================================
+// NCHW variants of the BN elementwise/stat kernels — used on the
+// default (NCHW, MIOpen-conv) path to replace torch's BatchNorm (5
+// MIOpen kernels + f32 promote/demote casts under autocast) and ReLU
+// with three cast-free bf16 kernels.  NCHW makes these EASIER than the
+// cl forms: one (b,c) plane is contiguous, so grid.y indexes the plane
+// and the channel is one modulo per block.
+// ===================================================================
+
+__global__ void ch_stats_nchw_kernel(
+    const cb_bf16* __restrict__ x, float* __restrict__ sums,
+    float* __restrict__ sumsq, int B, int C, long HW) {
+  __shared__ float red[4][2];
+  const int c = blockIdx.y;
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wv = tid >> 6;
+  float s = 0.f, q = 0.f;
+  const long step8 = (long)gridDim.x * blockDim.x * 8;
+  for (int b = 0; b < B; ++b) {
+    const cb_bf16* plane = x + ((long)b * C + c) * HW;
+    for (long i = ((long)blockIdx.x * blockDim.x + tid) * 8; i < HW;
+         i += step8) {
+      if (i + 8 <= HW) {
+        cb_bf16 v8[8];
+        *(cb_bf16x8*)v8 = *(const cb_bf16x8*)(plane + i);
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const float v = cb_b2f(v8[e]);
+          s += v;
+          q += v * v;
+        }
+      } else {
+        for (long j = i; j < HW; ++j) {
+          const float v = cb_b2f(plane[j]);
+          s += v;
+          q += v * v;
+        }
+      }
+    }
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    s += __shfl_down(s, off, WAVE);
+    q += __shfl_down(q, off, WAVE);
+  }
+  if (lane == 0) {
+    red[wv][0] = s;
+    red[wv][1] = q;
+  }
+  __syncthreads();
+  if (tid == 0) {
+    atomicAdd(sums + c, red[0][0] + red[1][0] + red[2][0] + red[3][0]);
+    atomicAdd(sumsq + c, red[0][1] + red[1][1] + red[2][1] + red[3][1]);
+  }
+}
+
+__global__ void bnrelu_apply_nchw_kernel(
+    const cb_bf16* __restrict__ x, const float* __restrict__ scale,
+    const float* __restrict__ shift, cb_bf16* __restrict__ xn,
+    int C, long HW) {
+  const int bc = blockIdx.y;
+  const int c = bc % C;
+  const float sc = scale[c], sh = shift[c];
+  const cb_bf16* plane = x + (long)bc * HW;
+  cb_bf16* out = xn + (long)bc * HW;
+  const long step8 = (long)gridDim.x * blockDim.x * 8;
+  for (long i = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 8; i < HW;
+       i += step8) {
+    if (i + 8 <= HW) {
+      cb_bf16 v8[8];
+      *(cb_bf16x8*)v8 = *(const cb_bf16x8*)(plane + i);
+#pragma unroll
+      for (int e = 0; e < 8; ++e)
+        v8[e] = cb_f2b(fmaxf(cb_b2f(v8[e]) * sc + sh, 0.f));
+      *(cb_bf16x8*)(out + i) = *(cb_bf16x8*)v8;
+    } else {
+      for (long j = i; j < HW; ++j)
+        out[j] = cb_f2b(fmaxf(cb_b2f(plane[j]) * sc + sh, 0.f));
+    }
+  }
+}
+
+__global__ void relu_mask_reduce_nchw_kernel(
+    const cb_bf16* __restrict__ dxn, const cb_bf16* __restrict__ x,
+    const float* __restrict__ scale, const float* __restrict__ shift,
+    const float* __restrict__ mean, const float* __restrict__ invstd,
+    cb_bf16* __restrict__ g, float* __restrict__ dgamma,
+    float* __restrict__ dbeta, int C, long HW) {
+  __shared__ float red[4][2];
+  const int bc = blockIdx.y;
+  const int c = bc % C;
+  const float sc = scale[c], sh = shift[c], mn = mean[c], is = invstd[c];
+  const cb_bf16* dplane = dxn + (long)bc * HW;
+  const cb_bf16* xplane = x + (long)bc * HW;
+  cb_bf16* gplane = g + (long)bc * HW;
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wv = tid >> 6;
+  float db = 0.f, dg = 0.f;
+  const long step8 = (long)gridDim.x * blockDim.x * 8;
+  for (long i = ((long)blockIdx.x * blockDim.x + tid) * 8; i < HW;
+       i += step8) {
+    const long hi = min(i + 8, HW);
+    for (long j = i; j < hi; ++j) {
+      const float xv = cb_b2f(xplane[j]);
+      const float gg = (xv * sc + sh) > 0.f ? cb_b2f(dplane[j]) : 0.f;
+      gplane[j] = cb_f2b(gg);
+      db += gg;
+      dg += gg * (xv - mn) * is;
+    }
+  }
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    db += __shfl_down(db, off, WAVE);
+    dg += __shfl_down(dg, off, WAVE);
+  }
+  if (lane == 0) {
+    red[wv][0] = db;
+    red[wv][1] = dg;
+  }
+  __syncthreads();
+  if (tid == 0) {
+    atomicAdd(dbeta + c, red[0][0] + red[1][0] + red[2][0] + red[3][0]);
+    atomicAdd(dgamma + c, red[0][1] + red[1][1] + red[2][1] + red[3][1]);
+  }
+}
+
+__global__ void bn_bwd_dx_nchw_kernel(
+    const cb_bf16* __restrict__ g, const cb_bf16* __restrict__ x,
+    const float* __restrict__ gamma, const float* __restrict__ invstd,
+    const float* __restrict__ mean, const float* __restrict__ dgamma,
+    const float* __restrict__ dbeta, float inv_n, cb_bf16* __restrict__ dx,
+    int C, long HW, int training) {
+  const int bc = blockIdx.y;
+  const int c = bc % C;
+  const float is = invstd[c];
+  const float gs = gamma[c] * is;
+  const float mn = mean[c];
+  const float dbn = training ? dbeta[c] * inv_n : 0.f;
+  const float dgn = training ? dgamma[c] * inv_n : 0.f;
+  const cb_bf16* gplane = g + (long)bc * HW;
+  const cb_bf16* xplane = x + (long)bc * HW;
+  cb_bf16* oplane = dx + (long)bc * HW;
+  const long step8 = (long)gridDim.x * blockDim.x * 8;
+  for (long i = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 8; i < HW;
+       i += step8) {
+    if (i + 8 <= HW) {
+      cb_bf16 g8[8], x8[8], o8[8];
+      *(cb_bf16x8*)g8 = *(const cb_bf16x8*)(gplane + i);
+      *(cb_bf16x8*)x8 = *(const cb_bf16x8*)(xplane + i);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        const float xhat = (cb_b2f(x8[e]) - mn) * is;
+        o8[e] = cb_f2b(gs * (cb_b2f(g8[e]) - dbn - xhat * dgn));
+      }
+      *(cb_bf16x8*)(oplane + i) = *(cb_bf16x8*)o8;
+    } else {
+      for (long j = i; j < HW; ++j) {
+        const float xhat = (cb_b2f(xplane[j]) - mn) * is;
+        oplane[j] = cb_f2b(gs * (cb_b2f(gplane[j]) - dbn - xhat * dgn));
+      }
+    }
+  }
+}
+
+// ------------------------------------------------------------ hosts
+std::vector<torch::Tensor> ch_stats_nchw(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() &&
+              x.scalar_type() == at::kBFloat16);
+  const int B = x.size(0), C = x.size(1);
+  const long HW = (long)x.size(2) * x.size(3);
+  auto buf = torch::empty({2, C}, x.options().dtype(at::kFloat));
+  cb_zero(buf);
+  auto sums = buf.select(0, 0);
+  auto sumsq = buf.select(0, 1);
+  const int pch = (int)std::min((HW + 2047) / 2048, (long)32);
+  hipLaunchKernelGGL(ch_stats_nchw_kernel, dim3(pch, C), dim3(256), 0,
+                     cb_stream(), (const cb_bf16*)x.data_ptr(),
+                     (float*)sums.data_ptr(), (float*)sumsq.data_ptr(),
+                     B, C, HW);
+  CB_OK("ch_stats_nchw");
+  return {sums, sumsq};
+}
+
+torch::Tensor bnrelu_apply_nchw(
+    torch::Tensor x, torch::Tensor scale, torch::Tensor shift) {
+  TORCH_CHECK(x.is_contiguous() && x.scalar_type() == at::kBFloat16);
+  const int B = x.size(0), C = x.size(1);
+  const long HW = (long)x.size(2) * x.size(3);
+  auto xn = at::empty_like(x);
+  const int pch = (int)std::min((HW + 2047) / 2048, (long)16);
+  hipLaunchKernelGGL(bnrelu_apply_nchw_kernel, dim3(pch, B * C), dim3(256),
+                     0, cb_stream(), (const cb_bf16*)x.data_ptr(),
+                     (const float*)scale.data_ptr(),
+                     (const float*)shift.data_ptr(),
+                     (cb_bf16*)xn.data_ptr(), C, HW);
+  CB_OK("bnrelu_apply_nchw");
+  return xn;
+}
+
+std::vector<torch::Tensor> relu_mask_reduce_nchw(
+    torch::Tensor dxn, torch::Tensor x, torch::Tensor scale,
+    torch::Tensor shift, torch::Tensor mean, torch::Tensor invstd) {
+  TORCH_CHECK(dxn.is_contiguous() && x.is_contiguous());
+  const int B = x.size(0), C = x.size(1);
+  const long HW = (long)x.size(2) * x.size(3);
+  auto g = at::empty_like(x);
+  auto gb = torch::empty({2, (long)C}, scale.options());
+  cb_zero(gb);
+  auto dgamma = gb.select(0, 0);
+  auto dbeta = gb.select(0, 1);
+  const int pch = (int)std::min((HW + 2047) / 2048, (long)16);
+  hipLaunchKernelGGL(relu_mask_reduce_nchw_kernel, dim3(pch, B * C),
+                     dim3(256), 0, cb_stream(),
+                     (const cb_bf16*)dxn.data_ptr(),
+                     (const cb_bf16*)x.data_ptr(),
+                     (const float*)scale.data_ptr(),
+                     (const float*)shift.data_ptr(),
+                     (const float*)mean.data_ptr(),
+                     (const float*)invstd.data_ptr(),
+                     (cb_bf16*)g.data_ptr(), (float*)dgamma.data_ptr(),
+                     (float*)dbeta.data_ptr(), C, HW);
+  CB_OK("relu_mask_reduce_nchw");
+  return {g, dgamma, dbeta};
+}
+
+torch::Tensor bn_bwd_dx_nchw(
+    torch::Tensor g, torch::Tensor x, torch::Tensor gamma,
+    torch::Tensor invstd, torch::Tensor mean, torch::Tensor dgamma,
+    torch::Tensor dbeta, double n, bool training) {
+  TORCH_CHECK(g.is_contiguous() && x.is_contiguous());
+  const int B = x.size(0), C = x.size(1);
+  const long HW = (long)x.size(2) * x.size(3);
+  auto dx = at::empty_like(x);
+  const int pch = (int)std::min((HW + 2047) / 2048, (long)16);
+  hipLaunchKernelGGL(bn_bwd_dx_nchw_kernel, dim3(pch, B * C), dim3(256), 0,
+                     cb_stream(),
+                     (const cb_bf16*)g.data_ptr(),
+                     (const cb_bf16*)x.data_ptr(),
+                     (const float*)gamma.data_ptr(),
+                     (const float*)invstd.data_ptr(),
+                     (const float*)mean.data_ptr(),
+                     (const float*)dgamma.data_ptr(),
+                     (const float*)dbeta.data_ptr(),
+                     (float)(1.0 / n), (cb_bf16*)dx.data_ptr(),
+                     C, HW, training ? 1 : 0);
+  CB_OK("bn_bwd_dx_nchw");
+  return dx;
+}

@@ -882,6 +882,16 @@ torch::Tensor bnrelu_apply(
 std::vector<torch::Tensor> relu_mask_reduce(
     torch::Tensor dxn, torch::Tensor x, torch::Tensor scale,
     torch::Tensor shift, torch::Tensor mean, torch::Tensor invstd);
+std::vector<torch::Tensor> ch_stats_nchw(torch::Tensor x);
+torch::Tensor bnrelu_apply_nchw(
+    torch::Tensor x, torch::Tensor scale, torch::Tensor shift);
+std::vector<torch::Tensor> relu_mask_reduce_nchw(
+    torch::Tensor dxn, torch::Tensor x, torch::Tensor scale,
+    torch::Tensor shift, torch::Tensor mean, torch::Tensor invstd);
+torch::Tensor bn_bwd_dx_nchw(
+    torch::Tensor g, torch::Tensor x, torch::Tensor gamma,
+    torch::Tensor invstd, torch::Tensor mean, torch::Tensor dgamma,
+    torch::Tensor dbeta, double n, bool training);
 torch::Tensor bn_bwd_dx(
     torch::Tensor g, torch::Tensor x, torch::Tensor gamma,
     torch::Tensor invstd, torch::Tensor mean, torch::Tensor dgamma,
@@ -922,6 +932,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bnconv_dgrad", &bnconv_dgrad);
   m.def("bnrelu_apply", &bnrelu_apply);
   m.def("relu_mask_reduce", &relu_mask_reduce);
+  m.def("ch_stats_nchw", &ch_stats_nchw);
+  m.def("bnrelu_apply_nchw", &bnrelu_apply_nchw);
+  m.def("relu_mask_reduce_nchw", &relu_mask_reduce_nchw);
+  m.def("bn_bwd_dx_nchw", &bn_bwd_dx_nchw);
   m.def("bn_bwd_dx", &bn_bwd_dx);
   m.def("bnrelu_conv_fwd", &bnrelu_conv_fwd);
   m.def("bnrelu_conv_fwd_fp8", &bnrelu_conv_fwd_fp8);

@@ -72,6 +72,12 @@ class _Bottleneck(nn.Module):
             # the load stage (csrc/conv_mfma.hip)
             out = fused_conv.bnrelu_conv(x, self.conv1.weight, self.bn1)
             out = fused_conv.bnrelu_conv(out, self.conv2.weight, self.bn2)
+        elif fused_conv.can_fuse_bnrelu(x):
+            # BN+ReLU as cast-free hand kernels, convs stay MIOpen (the
+            # full fused-conv path is opt-in; this elementwise fusion
+            # wins on the default NCHW path)
+            out = self.conv1(fused_conv.bnrelu(x, self.bn1))
+            out = self.conv2(fused_conv.bnrelu(out, self.bn2))
         else:
             out = self.conv1(F.relu(self.bn1(x)))
             out = self.conv2(F.relu(self.bn2(out)))
@@ -94,6 +100,8 @@ class _Transition(nn.Module):
             # hand stats/mask kernels + rocBLAS GEMMs (K > 96; aten's
             # channels_last wrw here picked a ~5 ms CK kernel)
             out = fused_conv.bnrelu_conv1x1_big(x, self.conv.weight, self.bn)
+        elif fused_conv.can_fuse_bnrelu(x):
+            out = self.conv(fused_conv.bnrelu(x, self.bn))
         else:
             out = self.conv(F.relu(self.bn(x)))
         return F.avg_pool2d(out, 2, ceil_mode=True)
@@ -127,7 +135,11 @@ class DenseNetEncoder(nn.Module):
         h = conv_1ch(x, self.stem.weight, stride=2, padding=3)
         if x.is_contiguous(memory_format=torch.channels_last):
             h = h.to(memory_format=torch.channels_last)
-        h = F.relu(self.stem_bn(h))
+        from ..ops import fused_conv as _fc
+        if _fc.can_fuse_bnrelu(h):
+            h = _fc.bnrelu(h, self.stem_bn)
+        else:
+            h = F.relu(self.stem_bn(h))
         h = F.max_pool2d(h, 2, ceil_mode=True)   # stride 4 after stem+pool
         a = self.blocks(h)
         mask = downsample_mask(x_mask, self.stride)

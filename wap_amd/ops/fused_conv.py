@@ -241,3 +241,64 @@ def bnrelu_conv1x1_big(x: Tensor, weight: Tensor,
     return _BNReluLinear.apply(x, w, bn.weight, bn.bias, bn.running_mean,
                                bn.running_var, bn.eps, bn.momentum,
                                bn.training)
+
+
+class _BNRelu(torch.autograd.Function):
+    """relu(bn(x)) as three cast-free bf16 NCHW kernels.
+
+    Under autocast, torch.batch_norm promotes to f32: each BN costs 5
+    MIOpen kernels plus bf16->f32->bf16 casts, and ReLU two more — the
+    dominant share of the elementwise ocean in the step profile.  Here:
+    forward = ch_stats_nchw + bn_finalize + bnrelu_apply_nchw, backward =
+    relu_mask_reduce_nchw + bn_bwd_dx_nchw, all bf16 in/out with f32
+    stats.  The conv stays MIOpen (NCHW) — this fuses only the
+    elementwise/stat stack around it."""
+
+    @staticmethod
+    def forward(ctx, x: Tensor, gamma: Tensor, beta: Tensor,
+                running_mean: Tensor, running_var: Tensor, eps: float,
+                momentum: float, training: bool):
+        from . import ext
+        lib = ext.require()
+        C = x.shape[1]
+        if training:
+            sums, sumsq = lib.ch_stats_nchw(x)
+            n = x.numel() / C
+        else:
+            sums = sumsq = torch.empty(0)
+            n = 1.0
+        scale, shift, mean, invstd = lib.bn_finalize(
+            sums, sumsq, gamma.float(), beta.float(), running_mean,
+            running_var, float(n), eps, momentum if training else 0.0,
+            training)
+        xn = lib.bnrelu_apply_nchw(x, scale, shift)
+        ctx.save_for_backward(x, gamma, scale, shift, mean, invstd)
+        ctx.training = training
+        ctx.n = n
+        return xn
+
+    @staticmethod
+    def backward(ctx, gxn: Tensor):
+        from . import ext
+        lib = ext.require()
+        x, gamma, scale, shift, mean, invstd = ctx.saved_tensors
+        g, dgamma, dbeta = lib.relu_mask_reduce_nchw(
+            gxn.contiguous(), x, scale, shift, mean, invstd)
+        dx = lib.bn_bwd_dx_nchw(g, x, gamma.float(), invstd, mean, dgamma,
+                                dbeta, ctx.n, ctx.training)
+        return (dx, dgamma.to(gamma.dtype), dbeta.to(gamma.dtype),
+                None, None, None, None, None)
+
+
+def can_fuse_bnrelu(x: Tensor) -> bool:
+    import os
+    if os.environ.get("WAP_FUSED_BNRELU", "1") == "0":
+        return False
+    return (use_hip(x) and x.dtype == torch.bfloat16
+            and x.is_contiguous())
+
+
+def bnrelu(x: Tensor, bn: torch.nn.BatchNorm2d) -> Tensor:
+    """relu(bn(x)) with the fused NCHW kernels."""
+    return _BNRelu.apply(x, bn.weight, bn.bias, bn.running_mean,
+                         bn.running_var, bn.eps, bn.momentum, bn.training)
