@@ -244,15 +244,17 @@ def bnrelu_conv1x1_big(x: Tensor, weight: Tensor,
 
 
 class _BNRelu(torch.autograd.Function):
-    """relu(bn(x)) as three cast-free bf16 NCHW kernels.
+    """relu(bn(x)) as three cast-free bf16 kernels (NCHW or NHWC).
 
     Under autocast, torch.batch_norm promotes to f32: each BN costs 5
     MIOpen kernels plus bf16->f32->bf16 casts, and ReLU two more — the
     dominant share of the elementwise ocean in the step profile.  Here:
-    forward = ch_stats_nchw + bn_finalize + bnrelu_apply_nchw, backward =
-    relu_mask_reduce_nchw + bn_bwd_dx_nchw, all bf16 in/out with f32
-    stats.  The conv stays MIOpen (NCHW) — this fuses only the
-    elementwise/stat stack around it."""
+    forward = ch_stats + bn_finalize + bnrelu_apply, backward =
+    relu_mask_reduce + bn_bwd_dx, all bf16 in/out with f32 stats, with
+    layout-matched kernel variants (the encoder runs channels_last in
+    practice: a [B,1,H,W] input is layout-ambiguous and the stem promotes
+    downstream tensors to NHWC).  The conv stays MIOpen — this fuses only
+    the elementwise/stat stack around it."""
 
     @staticmethod
     def forward(ctx, x: Tensor, gamma: Tensor, beta: Tensor,
@@ -261,8 +263,11 @@ class _BNRelu(torch.autograd.Function):
         from . import ext
         lib = ext.require()
         C = x.shape[1]
+        nchw = x.is_contiguous()
+        stats = lib.ch_stats_nchw if nchw else lib.ch_stats
+        apply_ = lib.bnrelu_apply_nchw if nchw else lib.bnrelu_apply
         if training:
-            sums, sumsq = lib.ch_stats_nchw(x)
+            sums, sumsq = stats(x)
             n = x.numel() / C
         else:
             sums = sumsq = torch.empty(0)
@@ -271,10 +276,11 @@ class _BNRelu(torch.autograd.Function):
             sums, sumsq, gamma.float(), beta.float(), running_mean,
             running_var, float(n), eps, momentum if training else 0.0,
             training)
-        xn = lib.bnrelu_apply_nchw(x, scale, shift)
+        xn = apply_(x, scale, shift)
         ctx.save_for_backward(x, gamma, scale, shift, mean, invstd)
         ctx.training = training
         ctx.n = n
+        ctx.nchw = nchw
         return xn
 
     @staticmethod
@@ -282,10 +288,15 @@ class _BNRelu(torch.autograd.Function):
         from . import ext
         lib = ext.require()
         x, gamma, scale, shift, mean, invstd = ctx.saved_tensors
-        g, dgamma, dbeta = lib.relu_mask_reduce_nchw(
-            gxn.contiguous(), x, scale, shift, mean, invstd)
-        dx = lib.bn_bwd_dx_nchw(g, x, gamma.float(), invstd, mean, dgamma,
-                                dbeta, ctx.n, ctx.training)
+        if ctx.nchw:
+            gxn = gxn.contiguous()
+            mask_reduce, bwd_dx = lib.relu_mask_reduce_nchw, lib.bn_bwd_dx_nchw
+        else:
+            gxn = gxn.contiguous(memory_format=torch.channels_last)
+            mask_reduce, bwd_dx = lib.relu_mask_reduce, lib.bn_bwd_dx
+        g, dgamma, dbeta = mask_reduce(gxn, x, scale, shift, mean, invstd)
+        dx = bwd_dx(g, x, gamma.float(), invstd, mean, dgamma, dbeta,
+                    ctx.n, ctx.training)
         return (dx, dgamma.to(gamma.dtype), dbeta.to(gamma.dtype),
                 None, None, None, None, None)
 
@@ -295,10 +306,12 @@ def can_fuse_bnrelu(x: Tensor) -> bool:
     if os.environ.get("WAP_FUSED_BNRELU", "1") == "0":
         return False
     return (use_hip(x) and x.dtype == torch.bfloat16
-            and x.is_contiguous())
+            and x.shape[1] <= 768
+            and (x.is_contiguous()
+                 or x.is_contiguous(memory_format=torch.channels_last)))
 
 
 def bnrelu(x: Tensor, bn: torch.nn.BatchNorm2d) -> Tensor:
-    """relu(bn(x)) with the fused NCHW kernels."""
+    """relu(bn(x)) with the fused kernels (layout-matched)."""
     return _BNRelu.apply(x, bn.weight, bn.bias, bn.running_mean,
                          bn.running_var, bn.eps, bn.momentum, bn.training)
