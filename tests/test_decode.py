@@ -143,7 +143,9 @@ def test_banked_finished_hyp_survives_eviction():
     from wap_amd.runtime.decode import BeamState
     st = BeamState.__new__(BeamState)
     st.b, st.k = 1, 2
+    st._rec_cap = 0
     T = 3
+    st.n_rec = T
     # slot 0 dies at t=1 with a GOOD normalized score (-1.0/2 = -0.5); by the
     # end both surviving slots are live continuations with worse norm scores.
     st.tokens = [torch.tensor([[4, 5]]), torch.tensor([[EOS_ID, 6]]),
@@ -170,6 +172,8 @@ def test_live_hyp_beats_worse_banked():
     from wap_amd.runtime.decode import BeamState
     st = BeamState.__new__(BeamState)
     st.b, st.k = 1, 2
+    st._rec_cap = 0
+    st.n_rec = 2
     st.tokens = [torch.tensor([[4, 5]]), torch.tensor([[EOS_ID, 6]])]
     st.parents = [torch.tensor([[0, 0]]), torch.tensor([[0, 1]])]
     st.dieds = [torch.zeros(1, 2, dtype=torch.bool) for _ in range(2)]

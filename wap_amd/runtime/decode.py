@@ -126,10 +126,16 @@ class BeamState:
             "brow": (torch.arange(b, device=dev) * beam).unsqueeze(1),
         }
         self.S["scores"][:, 0] = 0.0
-        self.tokens: List[Tensor] = []     # [B, k] per step
-        self.parents: List[Tensor] = []    # [B, k] back-pointers
-        self.dieds: List[Tensor] = []      # [B, k] finished-here flags
-        self.dscores: List[Tensor] = []    # [B, k] scores at that step
+        # per-step records: list mode (append clones) until
+        # ensure_record(maxlen) switches to preallocated [T,B,k] buffers —
+        # no allocation and one batched D2H at extract() instead of 4
+        # clones + 4 transfers per decoded token
+        self.tokens = []     # [B, k] per step (or [T,B,k] buffer)
+        self.parents = []    # [B, k] back-pointers
+        self.dieds = []      # [B, k] finished-here flags
+        self.dscores = []    # [B, k] scores at that step
+        self.n_rec = 0
+        self._rec_cap = 0
         self.graph = None
 
     def reset_from(self, other: "BeamState") -> None:
@@ -145,18 +151,42 @@ class BeamState:
         S["alive"].fill_(True)
         S["died"].fill_(False)
         S["lengths"].zero_()
-        self.tokens, self.parents = [], []
-        self.dieds, self.dscores = [], []
+        self.n_rec = 0
+        if self._rec_cap == 0:
+            self.tokens, self.parents = [], []
+            self.dieds, self.dscores = [], []
 
     def step(self, fused: bool = True) -> None:
         _beam_step_(self.model, self.S, self.b, self.k, fused)
         self.record()
 
+    def ensure_record(self, maxlen: int) -> None:
+        """Switch per-step records to preallocated [maxlen,B,k] buffers."""
+        if self._rec_cap >= maxlen:
+            self.n_rec = 0
+            return
+        dev = self.S["token"].device
+        b, k = self.b, self.k
+        self.tokens = torch.empty(maxlen, b, k, dtype=torch.long, device=dev)
+        self.parents = torch.empty_like(self.tokens)
+        self.dieds = torch.empty(maxlen, b, k, dtype=torch.bool, device=dev)
+        self.dscores = torch.empty(maxlen, b, k, device=dev)
+        self._rec_cap = maxlen
+        self.n_rec = 0
+
     def record(self) -> None:
-        self.tokens.append(self.S["token"].clone())
-        self.parents.append(self.S["parent"].clone())
-        self.dieds.append(self.S["died"].clone())
-        self.dscores.append(self.S["scores"].clone())
+        if self._rec_cap:
+            t = self.n_rec
+            self.tokens[t].copy_(self.S["token"], non_blocking=True)
+            self.parents[t].copy_(self.S["parent"], non_blocking=True)
+            self.dieds[t].copy_(self.S["died"], non_blocking=True)
+            self.dscores[t].copy_(self.S["scores"], non_blocking=True)
+        else:
+            self.tokens.append(self.S["token"].clone())
+            self.parents.append(self.S["parent"].clone())
+            self.dieds.append(self.S["died"].clone())
+            self.dscores.append(self.S["scores"].clone())
+        self.n_rec += 1
 
     # ---- hipGraph capture -------------------------------------------------
     def capture(self, fused: bool = True, autocast_bf16: bool = False) -> None:
@@ -194,14 +224,20 @@ class BeamState:
         by length-normalized score.  (Surviving dead slots are a subset of
         their own death events.)"""
         b = self.b
-        tokens = [t.cpu() for t in self.tokens]
-        parents = [p.cpu() for p in self.parents]
-        dieds = [d.cpu() for d in self.dieds]
-        dscores = [s.cpu() for s in self.dscores]
+        T = self.n_rec
+        if self._rec_cap:
+            tokens = self.tokens[:T].cpu()
+            parents = self.parents[:T].cpu()
+            dieds = self.dieds[:T].cpu()
+            dscores = self.dscores[:T].cpu()
+        else:
+            tokens = [t.cpu() for t in self.tokens]
+            parents = [p.cpu() for p in self.parents]
+            dieds = [d.cpu() for d in self.dieds]
+            dscores = [s.cpu() for s in self.dscores]
         live_norm = (self.S["scores"]
                      / self.S["lengths"].clamp_min(1).float()).cpu()
         alive = self.S["alive"].cpu()
-        T = len(tokens)
         out: List[List[int]] = []
         for i in range(b):
             best = None  # (norm_score, step, slot)
@@ -272,6 +308,7 @@ def beam_decode(
     state = BeamState(model, x, x_mask, beam)
     if use_graph is None:
         use_graph = x.is_cuda
+    state.ensure_record(maxlen)
     if use_graph and x.is_cuda:
         autocast_on = torch.is_autocast_enabled("cuda")
         key = _graph_key(model, state, fused, autocast_on)
@@ -286,6 +323,7 @@ def beam_decode(
         else:
             cached.reset_from(state)
             state = cached
+            state.ensure_record(maxlen)
         for t in range(maxlen):
             state.replay()
             if (t + 1) % sync_every == 0 and state.all_dead():
