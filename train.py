@@ -91,7 +91,7 @@ def validate(model, buckets, vocab, cfg, device, autocast_ctx) -> float:
             with autocast_ctx():
                 hyps = greedy_decode(
                     model, batch["x"], batch["x_mask"],
-                    maxlen=cfg.decode_maxlen, fused=True,
+                    maxlen=cfg.decode_maxlen, fused=False,
                     use_graph=on_gpu,
                 )
             refs = batch["y"]

@@ -299,11 +299,16 @@ def beam_decode(
     x_mask: Tensor,
     beam: int = 10,
     maxlen: int = 200,
-    fused: bool = True,
+    fused: bool = False,
     use_graph: Optional[bool] = None,
     sync_every: int = 8,
 ) -> List[List[int]]:
-    """Beam-search decode a batch -> token id lists (no eos)."""
+    """Beam-search decode a batch -> token id lists (no eos).
+
+    ``fused=True`` routes the GRU cells through the single-launch VALU
+    GEMV kernel — an r01 launch-count optimization that hipGraph replay
+    made obsolete: under graphs the rocBLAS gate-GEMM path (default)
+    measured 18.9k vs 7.9k tok/s (PERF.md r02)."""
     model.eval()
     state = BeamState(model, x, x_mask, beam)
     if use_graph is None:
