@@ -2,6 +2,9 @@
 """Minimal repro: MultiShapeGraphedTrainer over cycling padded shapes."""
 import os
 import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
 import torch
 
 from wap_amd.models import WAPModel
