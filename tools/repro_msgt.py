@@ -18,8 +18,14 @@ def main():
     ddp = BucketedDDP(m)
     opt = torch.optim.Adadelta(m.parameters(), lr=1.0, weight_decay=1e-4)
     tr = MultiShapeGraphedTrainer(m, ddp, opt, clip=100.0, autocast_bf16=True)
-    shapes = [(64, 128, 16), (64, 192, 16), (128, 128, 32), (64, 64, 16),
-              (128, 192, 16)]
+    # the real padded-shape set of a --synthetic 256 run (T up to 64,
+    # feature grids up to 12x48)
+    shapes = [(128, 320, 64), (128, 512, 64), (128, 640, 48),
+              (192, 320, 48), (192, 448, 32), (192, 448, 64),
+              (192, 512, 48), (192, 512, 64), (192, 576, 32),
+              (192, 576, 48), (192, 640, 64), (192, 704, 48),
+              (192, 704, 64), (192, 768, 32), (192, 768, 48),
+              (192, 768, 64)]
     nsteps = int(sys.argv[1]) if len(sys.argv) > 1 else 60
     for step in range(nsteps):
         h, w, t = shapes[step % len(shapes)]
