@@ -161,6 +161,14 @@ def main() -> None:
         trainer = MultiShapeGraphedTrainer(
             model, ddp, optimizer, clip=cfg.grad_clip, autocast_bf16=use_bf16
         )
+    if trainer is not None:
+        # The pinned-memory prefetch thread issues H2D copies while the
+        # trainer may be mid-hipGraph-capture on the main thread; a
+        # variable-shape training run produced a GPU memory-access fault
+        # with the two interleaved (r02).  Graph capture dominates the
+        # steady state here anyway, so the loader stays synchronous when
+        # graphs are on.
+        os.environ.setdefault("WAP_NO_PREFETCH", "1")
     loader = BucketLoader(
         train_buckets, rank=rank, world_size=world, seed=cfg.seed,
         device=device, pad_to=pad_to,
