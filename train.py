@@ -75,8 +75,16 @@ def load_split(cfg: WAPConfig, vocab: Vocab, split: str, seed: int):
     return buckets, dropped, samples
 
 
-def validate(model, buckets, vocab, cfg, device, autocast_ctx) -> float:
-    """Greedy-decode the validation buckets -> WER (rank 0 only)."""
+def validate(model, buckets, vocab, cfg, device, autocast_ctx,
+             use_graph: bool = True) -> float:
+    """Greedy-decode the validation buckets -> WER (rank 0 only).
+
+    ``use_graph``: capture/replay the decode step per shape.  Callers
+    pass False while a graphed TRAINER holds captured train-step graphs —
+    capturing decode graphs on top of them produced a GPU memory-access
+    fault on ROCm 7.x (r02, reproduced: epoch trains clean, fault appears
+    exactly at the first validation capture; decode graphs alone and
+    train graphs alone are both stable)."""
     pairs = []
     on_gpu = torch.device(device).type == "cuda"
     # pad validation buckets onto the SAME shape grid training uses,
@@ -92,7 +100,7 @@ def validate(model, buckets, vocab, cfg, device, autocast_ctx) -> float:
                 hyps = greedy_decode(
                     model, batch["x"], batch["x_mask"],
                     maxlen=cfg.decode_maxlen, fused=False,
-                    use_graph=on_gpu,
+                    use_graph=on_gpu and use_graph,
                 )
             refs = batch["y"]
             masks = batch["y_mask"]
@@ -213,7 +221,8 @@ def main() -> None:
             wer = math.nan
             if rank == 0:
                 model.eval()
-                res = validate(model, valid[0], vocab, cfg, device, autocast_ctx)
+                res = validate(model, valid[0], vocab, cfg, device,
+                               autocast_ctx, use_graph=(trainer is None))
                 wer = res.wer
                 is_best = wer < best_wer
                 best_wer = min(best_wer, wer)
