@@ -585,5 +585,9 @@ def test_manual_bptt_matches_autograd_gpu():
     assert _rel_err(lg1, lg0) < 2e-2, _rel_err(lg1, lg0)
     assert set(g1) == set(g0)
     for n in g0:
-        err = _rel_err(g1[n], g0[n])
-        assert err < 8e-2, (n, err)
+        # absolute floor: grads whose reference norm is cancellation-level
+        # (e.g. U_a.bias, ~1e-7) have no meaningful relative error between
+        # bf16 per-step accumulation and f32 batched accumulation
+        diff = (g1[n].float() - g0[n].float()).norm().item()
+        ref = g0[n].float().norm().item()
+        assert diff <= 8e-2 * ref + 1e-5, (n, diff, ref)
