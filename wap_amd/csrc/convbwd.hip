@@ -15,6 +15,8 @@
 //   bnconv_dgrad   dxn = conv_transpose(g_out, w) as an MFMA implicit
 //                  GEMM, fused with the relu mask (recomputed from x) and
 //                  the BN reductions dbeta = sum g, dgamma = sum g*xhat
+//   relu_mask_reduce  the same mask+reductions standalone (for convs
+//                  whose data-grad is a plain GEMM / MIOpen)
 //   bnrelu_apply   materialize xn = relu(bn(x)) once for the backward-side
 //                  wrw GEMM (rocBLAS 1x1 / MIOpen 3x3 — measured faster
 //                  than a hand LDS-transposed wrw kernel, see PERF.md)

@@ -1,16 +1,22 @@
-"""Fused pre-activation BN + ReLU + conv for DenseNet bottlenecks.
+"""Fused pre-activation BN + ReLU + conv paths for the DenseNet encoder.
 
-Forward runs the hand-written MFMA implicit-GEMM kernel
-(csrc/conv_mfma.hip): the BN affine + ReLU are applied at the input load
-stage, so the normalized tensor is never materialized (x is read once by
-the conv instead of BN-write + conv-read).
+All numerically verified on GPU against the fp32 eager composition, all
+OPT-IN after end-to-end A/B against MIOpen (PERF.md r02 tables):
 
-Backward recomputes the normalized input (cheap elementwise) and uses
-aten.convolution_backward for the conv grads + the standard batch-norm
-backward (train mode: batch-stat dependence included) in plain torch ops.
+- ``_BNReluConv`` (``WAP_FUSED_CONV=auto``): bottleneck convs — hand
+  batch-stats kernels + MFMA fused BN-ReLU-conv forward
+  (csrc/conv_mfma.hip) + MFMA dgrad fused with the relu mask / BN
+  reductions (csrc/convbwd.hip) + library wrw over a materialized xn +
+  hand BN backward.  ``=hybrid`` swaps the dgrad for MIOpen's.
+- ``_BNReluLinear`` (transitions, K > 96): hand stats/mask kernels +
+  rocBLAS GEMMs.
+- ``_BNRelu`` (``WAP_FUSED_BNRELU=1``): just the BN+ReLU elementwise
+  stack (layout-polymorphic), convs stay MIOpen.
+- fp8 / MX-scaled fp8 forward quantization helpers (BASELINE config 5 —
+  measured slower than bf16 at these shapes, kernels + parity tests kept).
 
-Constraints for the HIP path: bf16, channels_last, stride 1, odd square
-kernel, K_out <= 96 (DenseNet bottleneck convs are C->4k=96 and 96->k=24).
+Constraints for the conv kernels: bf16, channels_last, stride 1, odd
+square kernel, K_out <= 96 (bottlenecks are C->4k=96 and 96->k=24).
 """
 
 from __future__ import annotations
@@ -30,11 +36,13 @@ class _BNReluConv(torch.autograd.Function):
               replacing torch var_mean and the running-stat update ops)
               -> bnrelu_conv_fwd (MFMA implicit GEMM, BN+ReLU in the load)
     backward: bnconv_dgrad (MFMA conv-transpose fused with the relu mask
-              and the dbeta/dgamma reductions) + conv_wrw (MFMA, split-K
-              atomics) + bn_bwd_dx (one elementwise kernel)
+              and the dbeta/dgamma reductions) + library wrw over a
+              materialized xn + bn_bwd_dx (one elementwise kernel);
+              WAP_FUSED_CONV=hybrid uses MIOpen for both conv grads.
 
-    No aten.convolution_backward, no f32 recompute, no autocast casts —
-    the round-1 wrapper overhead that kept this op default-off (PERF.md).
+    No torch var_mean, no f32 recompute — the round-1 wrapper overhead is
+    gone; what keeps this opt-in now is the dgrad kernel itself losing to
+    MIOpen igemm_bwd e2e (PERF.md r02).
     """
 
     @staticmethod
