@@ -166,3 +166,31 @@ def test_vocab_oov_raises(vocab):
     import pytest as _pytest
     with _pytest.raises(KeyError):
         vocab.encode("definitely_not_a_token")
+
+
+def test_gen_pkl_cli_roundtrip(tmp_path):
+    """gen_pkl CLI: image dir (binary PGM) -> {uid: uint8[H,W]} pickle."""
+    import subprocess
+    import sys
+    d = tmp_path / "imgs"
+    d.mkdir()
+    rng = np.random.default_rng(3)
+    imgs = {}
+    for uid in ("a_1", "b_2"):
+        h, w = int(rng.integers(10, 30)), int(rng.integers(10, 40))
+        arr = rng.integers(0, 256, (h, w)).astype(np.uint8)
+        imgs[uid] = arr
+        header = f"P5\n{w} {h}\n255\n".encode()
+        (d / f"{uid}.pgm").write_bytes(header + arr.tobytes())
+    out = tmp_path / "feat.pkl"
+    r = subprocess.run(
+        [sys.executable, "-m", "wap_amd.data.gen_pkl", str(d), str(out)],
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+        capture_output=True, text=True)
+    assert r.returncode == 0, r.stderr
+    with open(out, "rb") as f:
+        feats = pickle.load(f)
+    assert set(feats) == set(imgs)
+    for uid in imgs:
+        assert feats[uid].dtype == np.uint8
+        assert (feats[uid] == imgs[uid]).all()
