@@ -21,8 +21,6 @@ square kernel, K_out <= 96 (bottlenecks are C->4k=96 and 96->k=24).
 
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 from torch import Tensor
 

@@ -25,7 +25,7 @@ the oracle tests).  Toggle with WAP_MANUAL_BPTT=0.
 
 from __future__ import annotations
 
-from typing import List, Optional, Tuple
+from typing import List
 
 import torch
 from torch import Tensor
