@@ -124,10 +124,16 @@ def main() -> None:
         # hipStreamSynchronize / hipDeviceSynchronize: on ROCm 7.x both of
         # those corrupt the next hipGraph replay's backward (measured:
         # grads explode exactly one step after every such sync; the
-        # memcpy-style sync .item() performs is safe).  The read drains the
-        # current stream, which carries all our work, so timing is exact.
+        # memcpy-style sync .item() performs is safe).  Rank alignment
+        # uses a dummy all-reduce on the current stream instead of
+        # dist.barrier() for the same reason: the barrier's internal wait
+        # may device-sync, while a stream collective + the D2H drain is
+        # exactly the (proven-safe) pattern the graphed trainer runs every
+        # step.  The read drains the current stream, which carries all our
+        # work, so timing is exact.
         if world > 1:
-            torch.distributed.barrier()
+            t = torch.ones(1, device=device if on_gpu else "cpu")
+            torch.distributed.all_reduce(t)
         if on_gpu and not os.environ.get("WAP_NO_BARRIER"):
             from wap_amd.runtime.graph_train import safe_stream_sync
             safe_stream_sync()
@@ -143,9 +149,12 @@ def main() -> None:
         loss = train_step()
         if rank == 0 and os.environ.get("WAP_LOSS_TRACE"):
             print(f"step[{i}] loss={float(loss.item()):.4f}", file=sys.stderr)
+    # read the loss BEFORE the closing barrier: .item() is itself the safe
+    # stream drain, and doing it first keeps the reported final_loss from
+    # any rank-alignment side effects
+    final_loss = float(loss.item())
     barrier_sync()
     elapsed = time.perf_counter() - t0
-    final_loss = float(loss.item())
 
     # MAX over ranks of the elapsed time (slowest rank defines throughput).
     if world > 1:
