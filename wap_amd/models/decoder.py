@@ -84,16 +84,19 @@ class Parser(nn.Module):
         mean_a = (a_flat * mask_flat.unsqueeze(-1)).sum(dim=1) / denom
         return torch.tanh(self.W_init(mean_a))
 
-    def readout(self, s: Tensor, c: Tensor, ey: Tensor) -> Tensor:
+    def readout(self, s: Tensor, c: Tensor, ey: Tensor,
+                w_cat: Optional[Tensor] = None) -> Tensor:
         """Maxout readout -> logits[..., K] (any leading dims).
 
         W_s s + W_c c + W_y ey == [s|c|ey] @ [W_s|W_c|W_y]^T: one GEMM
         instead of three + two adds.  The training unroll calls this ONCE
         over the stacked [B,T,...] states (the readout does not feed back
         into the recurrence), so its forward+backward GEMMs and the weight
-        concatenations happen once per sequence, not once per step."""
-        w_cat = torch.cat([self.W_s.weight, self.W_c.weight, self.W_y.weight],
-                          dim=1)
+        concatenations happen once per sequence, not once per step.  The
+        beam loop passes a hoisted ``w_cat`` (BeamState buffer)."""
+        if w_cat is None:
+            w_cat = torch.cat(
+                [self.W_s.weight, self.W_c.weight, self.W_y.weight], dim=1)
         x_cat = torch.cat([s, c.to(s.dtype), ey.to(s.dtype)], dim=-1)
         pre = torch.nn.functional.linear(x_cat, w_cat, self.W_s.bias)
         pre = pre.view(*pre.shape[:-1], self.readout_dim, self.maxout_pieces)
@@ -110,6 +113,7 @@ class Parser(nn.Module):
         fused: bool = False,
         w_cov: Optional[Tensor] = None,
         w_pack=None,
+        w_cat: Optional[Tensor] = None,
     ) -> Tuple[Tensor, Tensor, Tensor, Tensor]:
         """One decode step -> (logits, s_t, c_t, alpha)."""
         gru1 = self.gru1.step_fused if fused else self.gru1
@@ -118,7 +122,7 @@ class Parser(nn.Module):
         c_t, alpha = self.attention(s_hat, a_flat, ua, coverage, mask,
                                     w_cov, w_pack)
         s_t = gru2(c_t, s_hat)
-        logits = self.readout(s_t, c_t, y_prev_emb)
+        logits = self.readout(s_t, c_t, y_prev_emb, w_cat)
         return logits, s_t, c_t, alpha
 
     def forward(

@@ -51,14 +51,14 @@ def _beam_step_(model, S: Dict[str, Tensor], b: int, k: int, fused: bool) -> Non
     logits, s_new, _, alpha = m.step(
         S["emb"], S["s"], S["a_flat"], S["ua"], S["coverage"], S["mask"],
         fused=fused, w_cov=S["w_cov"], w_pack=(S["wpad"], S["wT"]),
+        w_cat=S["w_cat"],
     )
     logprob = torch.log_softmax(logits.float(), dim=-1).view(b, k, -1)
+    # Dead hypotheses may only emit <eos> at zero cost (score frozen);
+    # the frozen-row template is a static per-problem buffer.
     K = logprob.shape[-1]
-    # Dead hypotheses may only emit <eos> at zero cost (score frozen).
     dead = ~S["alive"]
-    frozen = torch.full_like(logprob, NEG_INF)
-    frozen[:, :, EOS_ID] = 0.0
-    logprob = torch.where(dead.unsqueeze(-1), frozen, logprob)
+    logprob = torch.where(dead.unsqueeze(-1), S["frozen"], logprob)
     total = S["scores"].unsqueeze(-1) + logprob               # [B,k,K]
     top_scores, flat_idx = total.view(b, k * K).topk(k, dim=1)
     parent = flat_idx // K                                    # [B,k]
@@ -103,10 +103,17 @@ class BeamState:
                  .detach().contiguous())
         from ..ops.functional import pack_coverage_weight
         wpad, wT = pack_coverage_weight(w_cov)
+        pm = model.parser
+        w_cat = torch.cat([pm.W_s.weight, pm.W_c.weight, pm.W_y.weight],
+                          dim=1).detach().contiguous()
+        frozen = torch.full((b, beam, pm.n_tokens), NEG_INF, device=dev)
+        frozen[:, :, EOS_ID] = 0.0
         self.S: Dict[str, Tensor] = {
             "w_cov": w_cov,
             "wpad": wpad,
             "wT": wT,
+            "w_cat": w_cat,
+            "frozen": frozen,
             "a_flat": a_flat_bk,
             "ua": _expand_for_beams(ua, beam).contiguous(),
             "mask": mask_bk.contiguous(),
@@ -144,7 +151,7 @@ class BeamState:
         state.  The hipGraph stays valid (same storage)."""
         S, O = self.S, other.S
         for n in ("a_flat", "ua", "mask", "s", "coverage", "emb", "w_cov",
-                  "wpad", "wT"):
+                  "wpad", "wT", "w_cat"):
             S[n].copy_(O[n])
         S["scores"].fill_(NEG_INF)
         S["scores"][:, 0] = 0.0
