@@ -167,3 +167,23 @@ def test_multishape_trainer_world4_alignment(tmp_path):
     port = 29737
     mp.spawn(_trainer_worker, args=(4, port, str(tmp_path)), nprocs=4,
              join=True)
+
+
+def test_bucket_clip_matches_torch_cpu():
+    """BucketedDDP.clip_grad_norm_ over flat buffers == torch's clip
+    (single process, CPU — the GPU twin lives in test_kernels_gpu)."""
+    import torch.nn as nn
+    from wap_amd.parallel import BucketedDDP
+    torch.manual_seed(51)
+    m1 = nn.Sequential(nn.Linear(32, 64), nn.Linear(64, 16))
+    m2 = nn.Sequential(nn.Linear(32, 64), nn.Linear(64, 16))
+    m2.load_state_dict(m1.state_dict())
+    x = torch.randn(4, 32)
+    ddp = BucketedDDP(m2, bucket_bytes=4096)  # multiple buckets
+    m1(x).square().sum().backward()
+    m2(x).square().sum().backward()
+    n1 = torch.nn.utils.clip_grad_norm_(m1.parameters(), 0.25)
+    n2 = ddp.clip_grad_norm_(0.25)
+    assert torch.allclose(n1, n2, rtol=1e-6)
+    for p1, p2 in zip(m1.parameters(), m2.parameters()):
+        assert torch.allclose(p1.grad, p2.grad, rtol=1e-6, atol=1e-8)
